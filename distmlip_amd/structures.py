@@ -78,10 +78,11 @@ def bcc_li(reps: int, a: float = 3.51, jitter: float = 0.1, seed: int = 0,
 
 def _cubic_supercell(basis, reps, a, jitter, seed, species_index) -> Structure:
     rng = np.random.default_rng(seed)
-    r = np.arange(reps)
-    cells = np.stack(np.meshgrid(r, r, r, indexing="ij"), axis=-1).reshape(-1, 3)
-    frac = ((cells[:, None, :] + basis[None, :, :]) / reps).reshape(-1, 3)
-    lattice = np.eye(3) * (a * reps)
+    reps3 = (reps, reps, reps) if np.isscalar(reps) else tuple(reps)
+    rs = [np.arange(r) for r in reps3]
+    cells = np.stack(np.meshgrid(*rs, indexing="ij"), axis=-1).reshape(-1, 3)
+    frac = ((cells[:, None, :] + basis[None, :, :]) / np.array(reps3)).reshape(-1, 3)
+    lattice = np.diag(a * np.array(reps3, dtype=float))
     if jitter > 0:
         frac = _jitter(frac, lattice, jitter, rng)
     n = len(frac)

@@ -1,0 +1,202 @@
+"""CPU oracle for the CHGNet energy+force forward (full graph, no partitions).
+
+An INDEPENDENT plain-torch restatement of the computation the reference
+orchestrates in implementations/matgl/models/chgnet.py:21-453 +
+implementations/matgl/pes.py:50-146, written straight-line over the full
+(unpartitioned) graph:
+
+  positions (chgnet.py:44-64) -> bond_vec/bond_dist (chgnet.py:96-100)
+  -> RBF * polynomial_cutoff applied to the RBF output (chgnet.py:115-124)
+  -> bond-graph geometry + theta + Fourier (chgnet.py:129-197)
+  -> embeddings (chgnet.py:231-251) -> shared weights (chgnet.py:272-294)
+  -> (n_blocks-1) x [atom conv; edge_to_bond; bond-node conv; bond_to_edge;
+     angle conv] (chgnet.py:296-368)
+  -> sitewise readout (chgnet.py:391-398) -> final atom block
+  (chgnet.py:400-419) -> final_layer + sum readout (chgnet.py:422-440)
+  -> scale/shift + element refs (pes.py:109-113) -> forces = -dE/dpos
+  (pes.py:121-124), stress via strain gradient (pes.py:140-145).
+
+Model-arithmetic parity vs upstream matgl is UNPINNED (oracle/__init__.py);
+this file IS the executable definition the product must match.  Tests use
+it in fp64 (tight partition-invariance) and fp32 (GPU parity tolerance).
+
+TEST INFRASTRUCTURE ONLY — see oracle/__init__.py.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from distmlip_amd.model import (
+    CHGNetCore,
+    bond_expansion_from_dist,
+    compute_theta,
+    fourier_expansion,
+)
+
+
+def build_full_line_graph(src: np.ndarray, dst: np.ndarray,
+                          within_idx: np.ndarray):
+    """Full-graph bond (line) graph.
+
+    Bond-graph NODES are the directed atom edges within the three-body
+    cutoff (`within_idx` into the global edge arrays).  Line EDGES connect
+    bond b1=(a->b) to bond b2=(b->c) whenever dst(b1)==src(b2) and
+    dst(b2)!=src(b1) (the backtrack skip, utils.c:727-729 — note the skip
+    compares ATOM ids only, so a->b->a via a different periodic image is
+    also skipped, a reference quirk we preserve).  The center atom of the
+    line is src(b2)=dst(b1)=b (utils.c:733).
+
+    Returns (line_src, line_dst, center_atom): int64 arrays of local bond
+    ids (indices into within_idx) and global atom ids.
+    """
+    b_src = np.asarray(src)[within_idx]
+    b_dst = np.asarray(dst)[within_idx]
+    nb = len(b_src)
+    order = np.argsort(b_src, kind="stable")         # bonds grouped by src atom
+    sorted_src = b_src[order]
+    n_atoms = int(max(b_src.max(initial=-1), b_dst.max(initial=-1)) + 1) if nb else 0
+    grp_start = np.searchsorted(sorted_src, np.arange(n_atoms), side="left")
+    grp_end = np.searchsorted(sorted_src, np.arange(n_atoms), side="right")
+
+    counts = (grp_end - grp_start)[b_dst] if nb else np.zeros(0, np.int64)
+    l_src = np.repeat(np.arange(nb, dtype=np.int64), counts)
+    # for each e1, the e2 candidates are order[grp_start[dst(e1)] : grp_end[...]]
+    offs = np.concatenate([np.arange(c) for c in counts]) if nb and counts.sum() else \
+        np.zeros(0, np.int64)
+    l_dst = order[np.repeat(grp_start[b_dst], counts) + offs] if nb else \
+        np.zeros(0, np.int64)
+    keep = b_dst[l_dst] != b_src[l_src]               # backtrack skip
+    l_src, l_dst = l_src[keep], l_dst[keep]
+    center = b_src[l_dst]
+    return l_src.astype(np.int64), l_dst.astype(np.int64), center.astype(np.int64)
+
+
+def oracle_forward(core: CHGNetCore, structure, src, dst, offsets, within_idx,
+                   dtype: torch.dtype = torch.float64,
+                   compute_forces: bool = True,
+                   compute_stress: bool = False):
+    """Full-graph CHGNet E+F forward on CPU.  Returns a dict."""
+    cfg = core.config
+    core = core.to(dtype)
+    lat0 = torch.tensor(np.asarray(structure.lattice), dtype=dtype)
+    strain = torch.zeros(3, 3, dtype=dtype)
+    if compute_stress:
+        strain.requires_grad_(True)
+    lattice = lat0 @ (torch.eye(3, dtype=dtype) + strain)   # chgnet.py:41-43
+
+    frac = torch.tensor(np.asarray(structure.frac_coords), dtype=dtype)
+    pos = frac @ lattice                                     # chgnet.py:58-61
+    if compute_forces:
+        pos = pos.detach().clone()
+        pos.requires_grad_(True)
+
+    species = torch.tensor(np.asarray(structure.species), dtype=torch.long)
+    src_t = torch.tensor(np.asarray(src), dtype=torch.long)
+    dst_t = torch.tensor(np.asarray(dst), dtype=torch.long)
+    off_t = torch.tensor(np.asarray(offsets), dtype=dtype)
+
+    offshift = off_t @ lattice                               # chgnet.py:55-57
+    bond_vec = pos[dst_t] + offshift - pos[src_t]            # chgnet.py:96-99
+    bond_dist = torch.linalg.norm(bond_vec, dim=1)           # chgnet.py:100
+
+    bond_expansion = bond_expansion_from_dist(
+        bond_dist, core.rbf_freq_atom, cfg.cutoff, cfg.cutoff_exponent)
+
+    v = core.atom_embedding(species)
+    e = core.bond_embedding(bond_expansion)
+
+    w_ab = core.atom_bond_weights(bond_expansion)
+    w_bb = core.bond_bond_weights(bond_expansion)
+
+    use_bg = cfg.use_bond_graph and len(within_idx) > 0
+    if use_bg:
+        within_t = torch.tensor(np.asarray(within_idx), dtype=torch.long)
+        l_src_np, l_dst_np, center_np = build_full_line_graph(src, dst, within_idx)
+        l_src = torch.tensor(l_src_np, dtype=torch.long)
+        l_dst = torch.tensor(l_dst_np, dtype=torch.long)
+        center = torch.tensor(center_np, dtype=torch.long)
+
+        bond_expansion3 = bond_expansion_from_dist(
+            bond_dist[within_t], core.rbf_freq_bond, cfg.three_body_cutoff,
+            cfg.cutoff_exponent)
+        w_3b = core.threebody_bond_weights(bond_expansion3)
+
+        bv_b = bond_vec[within_t]
+        theta = compute_theta(bv_b[l_src], bv_b[l_dst])      # chgnet.py:190-194
+        a = core.angle_embedding(fourier_expansion(theta, core.angle_freq))
+        n = e[within_t]                                      # chgnet.py:255-261
+
+    n_atoms = len(species)
+
+    def atom_conv(block, v, e):
+        """edge update then node update (node sees updated edges)."""
+        x = torch.cat([v[src_t], v[dst_t], e], dim=1)
+        e = e + block.edge_mlp(x) * w_bb
+        x = torch.cat([v[src_t], v[dst_t], e], dim=1)
+        msg = block.node_mlp(x) * w_ab
+        v = v + torch.zeros_like(v).index_add_(0, dst_t, msg)
+        return v, e
+
+    for layer_i in range(cfg.n_blocks - 1):                  # chgnet.py:296-368
+        v, e = atom_conv(core.atom_convs[layer_i], v, e)
+        if use_bg:
+            n = e[within_t]                                  # edge_to_bond
+            blk = core.bond_convs[layer_i]
+            x = torch.cat([n[l_src], n[l_dst], a, v[center]], dim=1)
+            msg = blk.bond_mlp(x) * w_3b[l_src]
+            n = n + torch.zeros_like(n).index_add_(0, l_dst, msg)
+            e = e.index_put((within_t,), n)                  # bond_to_edge
+            x = torch.cat([n[l_src], n[l_dst], a, v[center]], dim=1)
+            a = a + blk.angle_mlp(x)
+
+    site_props = core.sitewise_readout(v)                    # chgnet.py:391-398
+
+    v, e = atom_conv(core.atom_convs[-1], v, e)              # chgnet.py:400-419
+
+    atom_energies = core.final_layer(v)                      # chgnet.py:422-440
+    total_e = atom_energies.sum()
+
+    total_e = core.data_std * total_e + core.data_mean       # pes.py:109
+    total_e = total_e + core.element_refs[species].sum()     # pes.py:111-113
+
+    out = {"energy": total_e, "site_props": site_props,
+           "atom_energies": atom_energies.detach()}
+    if compute_forces:
+        grads = [pos, strain] if compute_stress else [pos]
+        gv = torch.autograd.grad(total_e, grads, retain_graph=False)
+        out["forces"] = -gv[0]                               # pes.py:121-124
+        if compute_stress:
+            volume = abs(np.linalg.det(np.asarray(structure.lattice)))
+            out["stress"] = -gv[1] / volume * -160.21766208  # pes.py:140-145
+    return out
+
+
+# ---------------------------------------------------------------------------
+# CPU ops backend for the DISTRIBUTED orchestration (test injection only).
+# Implements the distmlip_amd.ops_base.OpsBackend protocol in plain torch so
+# the product's partition/halo orchestration can run on CPU in tests.  The
+# product default backend is the HIP one, which refuses to run without the
+# extension — this class must never be reachable from product code.
+# ---------------------------------------------------------------------------
+
+from distmlip_amd.ops_base import ComposedMixin
+
+
+class CpuRefOps(ComposedMixin):
+    """Plain-torch implementation of the ops protocol (tests only)."""
+
+    is_reference = True
+
+    def gather(self, x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+        return x[idx]
+
+    def scatter_sum(self, msg: torch.Tensor, idx: torch.Tensor, n_out: int) -> torch.Tensor:
+        out = torch.zeros((n_out,) + tuple(msg.shape[1:]), dtype=msg.dtype,
+                          device=msg.device)
+        return out.index_add(0, idx, msg)
+
+    def edge_geom(self, pos: torch.Tensor, src: torch.Tensor, dst: torch.Tensor,
+                  offshift: torch.Tensor):
+        bv = pos[dst] + offshift - pos[src]
+        return bv, torch.linalg.norm(bv, dim=1)
