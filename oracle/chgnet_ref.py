@@ -88,8 +88,12 @@ def oracle_forward(core: CHGNetCore, structure, src, dst, offsets, within_idx,
     frac = torch.tensor(np.asarray(structure.frac_coords), dtype=dtype)
     pos = frac @ lattice                                     # chgnet.py:58-61
     if compute_forces:
-        pos = pos.detach().clone()
-        pos.requires_grad_(True)
+        # reference keeps positions CONNECTED to strain (chgnet.py:63-64:
+        # requires_grad_ + retain_grad on the non-leaf product), so the
+        # stress gradient flows through both positions and offshift
+        if not pos.requires_grad:
+            pos.requires_grad_(True)
+        pos.retain_grad()
 
     species = torch.tensor(np.asarray(structure.species), dtype=torch.long)
     src_t = torch.tensor(np.asarray(src), dtype=torch.long)

@@ -1,0 +1,41 @@
+import os
+import sys
+
+import numpy as np
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: needs a HIP GPU (run on MI355X)")
+
+
+@pytest.fixture(scope="session")
+def ref_graph_backend():
+    """The reference's own compiled C module (exact graph oracle)."""
+    from oracle import refmod
+    if not refmod.available() and not refmod.build_if_possible():
+        pytest.skip("reference graph oracle not built (oracle/_ref)")
+    return refmod.get_subgraphs_fast
+
+
+@pytest.fixture(scope="session")
+def core64():
+    from distmlip_amd.model import CHGNetCore
+    return CHGNetCore.seeded(seed=0).double()
+
+
+@pytest.fixture(scope="session")
+def si_slab():
+    """Elongated Si cell valid for 2-3 partitions (width check utils.c:1512)."""
+    from distmlip_amd.structures import diamond_si
+    return diamond_si((12, 2, 2), jitter=0.12, seed=2)
+
+
+@pytest.fixture(scope="session")
+def si_slab_graph(si_slab):
+    from oracle.graph_ref import brute_force_neighbors
+    return brute_force_neighbors(si_slab.frac_coords, si_slab.lattice,
+                                 si_slab.pbc, 6.0, 3.0)
