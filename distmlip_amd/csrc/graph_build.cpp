@@ -263,11 +263,17 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
     // --- partition rule (utils.c:1370-1456)
     Walls walls;
     if (P > 1) {
+        // QUIRK preserved: the reference picks the partition dimension from
+        // "cartesian" coordinates computed as cart = L . frac with lattice
+        // ROWS as the matrix rows (fast.c:16-27 fractional_to_cartesian),
+        // i.e. the TRANSPOSE of the convention FPIS/ase use (cart = frac @ L).
+        // Identical for diagonal lattices, different for skewed cells.
         double cmin[3] = {1e300, 1e300, 1e300}, cmax[3] = {-1e300, -1e300, -1e300};
         for (int64_t i = 0; i < n; ++i) {
             const double u = frac[3 * i], v = frac[3 * i + 1], w = frac[3 * i + 2];
             double c[3];
-            for (int k = 0; k < 3; ++k) c[k] = u * lat[0 + k] + v * lat[3 + k] + w * lat[6 + k];
+            for (int k = 0; k < 3; ++k)
+                c[k] = u * lat[3 * k] + v * lat[3 * k + 1] + w * lat[3 * k + 2];
             for (int k = 0; k < 3; ++k) { cmin[k] = std::min(cmin[k], c[k]); cmax[k] = std::max(cmax[k], c[k]); }
         }
         int dim = 0;
