@@ -72,7 +72,15 @@ class _PartView(ctypes.Structure):
                 ("line_src", POINTER(c_int64)), ("line_dst", POINTER(c_int64)),
                 ("line_center", POINTER(c_int64)),
                 ("map_de", POINTER(c_int64)), ("map_ude", POINTER(c_int64)),
-                ("bde_edge_gids", POINTER(c_int64))]
+                ("bde_edge_gids", POINTER(c_int64)),
+                ("row_ptr", POINTER(c_int64)),
+                ("src_perm", POINTER(c_int64)),
+                ("src_row_ptr", POINTER(c_int64)),
+                ("line_row_ptr", POINTER(c_int64)),
+                ("line_src_perm", POINTER(c_int64)),
+                ("line_src_row_ptr", POINTER(c_int64)),
+                ("center_perm", POINTER(c_int64)),
+                ("center_row_ptr", POINTER(c_int64))]
 
 
 class _Owner:
@@ -154,15 +162,27 @@ def build_graph(frac_coords, lattice, pbc, cutoff, bond_cutoff, tol,
             "bde_edge_gids": _as_np(pv.bde_edge_gids, pv.n_bonds, np.int64, owner),
             "n_bonds": pv.n_bonds,
             "n_owned_bonds": pv.n_owned_bonds,
+            "row_ptr": _as_np(pv.row_ptr, pv.n_nodes + 1, np.int64, owner),
+            "src_perm": _as_np(pv.src_perm, pv.n_edges, np.int64, owner),
+            "src_row_ptr": _as_np(pv.src_row_ptr, pv.n_nodes + 1, np.int64, owner),
+            "line_row_ptr": _as_np(pv.line_row_ptr, pv.n_bonds + 1, np.int64, owner),
+            "line_src_perm": _as_np(pv.line_src_perm, pv.n_lines, np.int64, owner),
+            "line_src_row_ptr": _as_np(pv.line_src_row_ptr, pv.n_bonds + 1,
+                                       np.int64, owner),
+            "center_perm": _as_np(pv.center_perm, pv.n_lines, np.int64, owner),
+            "center_row_ptr": _as_np(pv.center_row_ptr, pv.n_nodes + 1,
+                                     np.int64, owner),
         })
     return owner, g, parts
 
 
 def get_subgraphs_fast(cart_coords, cutoff, pbc, lattice, num_partitions,
                        bond_cutoff, tol, num_threads, use_bond_graph,
-                       frac_coords):
+                       frac_coords, return_csr=False):
     """Reference-compatible entry (subgraph_creation_fast.c:92-453 tuple
-    + one extension element: per-partition per-BDE global edge ids)."""
+    + one extension element: per-partition per-BDE global edge ids).
+    With return_csr=True also returns the per-partition CSR dicts the HIP
+    kernel path consumes."""
     _owner, g, parts = build_graph(frac_coords, lattice, pbc, cutoff,
                                    bond_cutoff, tol, num_partitions,
                                    num_threads, use_bond_graph)
@@ -172,7 +192,7 @@ def get_subgraphs_fast(cart_coords, cutoff, pbc, lattice, num_partitions,
 
     local_coords = [wrapped_cart[p["global_ids"]] for p in parts]
 
-    return (
+    ret = (
         [p["src_local"] for p in parts],
         [p["dst_local"] for p in parts],
         [p["markers"] for p in parts],
@@ -191,3 +211,6 @@ def get_subgraphs_fast(cart_coords, cutoff, pbc, lattice, num_partitions,
         [p["line_center"] for p in parts],
         [p["bde_edge_gids"] for p in parts],    # build extension
     )
+    if return_csr:
+        return ret, parts
+    return ret

@@ -59,8 +59,23 @@ struct Part {
     std::vector<int64_t> markers, line_markers, map_de, map_ude;
     rvec<int64_t> global_ids, src_local, dst_local, edge_gids;
     rvec<int64_t> line_src, line_dst, line_center, bde_edge_gids;
+    rvec<int64_t> row_ptr, src_perm, src_row_ptr;
+    rvec<int64_t> line_row_ptr, line_src_perm, line_src_row_ptr;
+    rvec<int64_t> center_perm, center_row_ptr;
     int64_t n_owned = 0, n_owned_bonds = 0;
 };
+
+// stable counting sort of [0..m) by key[i] in [0..nkeys); emits perm and
+// CSR row_ptr[nkeys+1]
+void counting_csr(const int64_t* key, int64_t m, int64_t nkeys,
+                  rvec<int64_t>& perm, rvec<int64_t>& rptr) {
+    rptr.assign(nkeys + 1, 0);
+    for (int64_t i = 0; i < m; ++i) ++rptr[key[i] + 1];
+    for (int64_t k = 0; k < nkeys; ++k) rptr[k + 1] += rptr[k];
+    perm.resize(m);
+    std::vector<int64_t> fill(nkeys, 0);
+    for (int64_t i = 0; i < m; ++i) perm[rptr[key[i]] + fill[key[i]]++] = i;
+}
 
 }  // namespace
 
@@ -386,9 +401,12 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
         for (int64_t i = 0; i < (int64_t)pt.global_ids.size(); ++i)
             g2l_node[pt.global_ids[i]] = i;
 
-        // local edges, ascending global edge id (count+fill over thread chunks)
+        // local edges: gather in global-id order (count+fill over thread
+        // chunks), then stable counting sort by dst_local -> DST-SORTED
+        // edge arrays + CSR row_ptr (the scatter-add layout)
         {
             const int T = nthreads;
+            const int64_t Nn = (int64_t)pt.global_ids.size();
             std::vector<int64_t> tcnt(T + 1, 0);
 #pragma omp parallel num_threads(T)
             {
@@ -400,7 +418,8 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
             }
             for (int t = 0; t < T; ++t) tcnt[t + 1] += tcnt[t];
             const int64_t Ep = tcnt[T];
-            pt.src_local.resize(Ep); pt.dst_local.resize(Ep); pt.edge_gids.resize(Ep);
+            rvec<int64_t> tsrc, tdst, tgid;
+            tsrc.resize(Ep); tdst.resize(Ep); tgid.resize(Ep);
 #pragma omp parallel num_threads(T)
             {
                 const int t = omp_get_thread_num();
@@ -408,13 +427,24 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
                 int64_t o = tcnt[t];
                 for (int64_t e = lo; e < hi; ++e)
                     if (eowner[e] == p) {
-                        pt.src_local[o] = g2l_node[g->src[e]];
-                        pt.dst_local[o] = g2l_node[g->dst[e]];
-                        pt.edge_gids[o] = e;
-                        if (use_bond_graph) g2l_edge[e] = o;
+                        tsrc[o] = g2l_node[g->src[e]];
+                        tdst[o] = g2l_node[g->dst[e]];
+                        tgid[o] = e;
                         ++o;
                     }
             }
+            rvec<int64_t> eperm;
+            counting_csr(tdst.data(), Ep, Nn, eperm, pt.row_ptr);
+            pt.src_local.resize(Ep); pt.dst_local.resize(Ep); pt.edge_gids.resize(Ep);
+#pragma omp parallel for num_threads(T) schedule(static)
+            for (int64_t i = 0; i < Ep; ++i) {
+                const int64_t j = eperm[i];
+                pt.src_local[i] = tsrc[j];
+                pt.dst_local[i] = tdst[j];
+                pt.edge_gids[i] = tgid[j];
+                if (use_bond_graph) g2l_edge[tgid[j]] = i;
+            }
+            counting_csr(pt.src_local.data(), Ep, Nn, pt.src_perm, pt.src_row_ptr);
         }
 
         if (!use_bond_graph) continue;
@@ -486,7 +516,8 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
         }
         for (int64_t b = 0; b < B; ++b) lcnt[b + 1] += lcnt[b];
         const int64_t L = lcnt[B];
-        pt.line_src.resize(L); pt.line_dst.resize(L); pt.line_center.resize(L);
+        rvec<int64_t> tls, tld, tlc;
+        tls.resize(L); tld.resize(L); tlc.resize(L);
 #pragma omp parallel for num_threads(nthreads) schedule(dynamic, 512)
         for (int64_t b1 = 0; b1 < B; ++b1) {
             const int64_t a2 = bdes[b1].dst_a;
@@ -495,13 +526,28 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
                 const int64_t b2 = by_src[q];
                 const BRec& e2 = bdes[b2];
                 if (e2.needs && e2.dst_a != bdes[b1].src_a) {
-                    pt.line_src[o] = b1;
-                    pt.line_dst[o] = b2;
-                    pt.line_center[o] = g2l_node[e2.src_a];        // utils.c:733,753-760
+                    tls[o] = b1;
+                    tld[o] = b2;
+                    tlc[o] = g2l_node[e2.src_a];                   // utils.c:733,753-760
                     ++o;
                 }
             }
         }
+        // l_dst-sorted line arrays + CSRs (scatter layouts for the kernels)
+        rvec<int64_t> lperm;
+        counting_csr(tld.data(), L, B, lperm, pt.line_row_ptr);
+        pt.line_src.resize(L); pt.line_dst.resize(L); pt.line_center.resize(L);
+#pragma omp parallel for num_threads(nthreads) schedule(static)
+        for (int64_t i = 0; i < L; ++i) {
+            const int64_t j = lperm[i];
+            pt.line_src[i] = tls[j];
+            pt.line_dst[i] = tld[j];
+            pt.line_center[i] = tlc[j];
+        }
+        counting_csr(pt.line_src.data(), L, B, pt.line_src_perm,
+                     pt.line_src_row_ptr);
+        counting_csr(pt.line_center.data(), L, (int64_t)pt.global_ids.size(),
+                     pt.center_perm, pt.center_row_ptr);
     }
     return 0;
 }
@@ -569,6 +615,14 @@ int dm_graph_partition_view(const dm_graph* g, int32_t partition,
     out->map_de = pt.map_de.data();
     out->map_ude = pt.map_ude.data();
     out->bde_edge_gids = pt.bde_edge_gids.data();
+    out->row_ptr = pt.row_ptr.data();
+    out->src_perm = pt.src_perm.data();
+    out->src_row_ptr = pt.src_row_ptr.data();
+    out->line_row_ptr = pt.line_row_ptr.empty() ? nullptr : pt.line_row_ptr.data();
+    out->line_src_perm = pt.line_src_perm.data();
+    out->line_src_row_ptr = pt.line_src_row_ptr.empty() ? nullptr : pt.line_src_row_ptr.data();
+    out->center_perm = pt.center_perm.data();
+    out->center_row_ptr = pt.center_row_ptr.empty() ? nullptr : pt.center_row_ptr.data();
     return 0;
 }
 

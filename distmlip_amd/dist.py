@@ -115,19 +115,29 @@ class Distributed:
         allowed: it yields a single partition with all nodes pure and no
         halo (the 1-GPU path).
         """
-        if backend is None:
-            from distmlip_amd import capi
-            backend = capi.get_subgraphs_fast
-
         cart_coords = np.ascontiguousarray(cart_coords, dtype=float)
         frac_coords = np.ascontiguousarray(frac_coords, dtype=float)
         lattice_matrix = np.ascontiguousarray(lattice_matrix, dtype=float)
 
-        out = backend(
-            cart_coords, float(cutoff), np.asarray(pbc, dtype=np.int64),
-            lattice_matrix, int(num_partitions), float(three_body_cutoff),
-            float(tol), int(num_threads), bool(use_bond_graph), frac_coords,
-        )
+        csr_parts = None
+        if backend is None:
+            from distmlip_amd import capi
+            backend = capi.get_subgraphs_fast
+            # native builder: keep the per-partition CSR layouts (scatter-add
+            # row pointers + permutation CSRs) for the HIP kernel path
+            out = backend(
+                cart_coords, float(cutoff), np.asarray(pbc, dtype=np.int64),
+                lattice_matrix, int(num_partitions), float(three_body_cutoff),
+                float(tol), int(num_threads), bool(use_bond_graph), frac_coords,
+                return_csr=True,
+            )
+            out, csr_parts = out
+        else:
+            out = backend(
+                cart_coords, float(cutoff), np.asarray(pbc, dtype=np.int64),
+                lattice_matrix, int(num_partitions), float(three_body_cutoff),
+                float(tol), int(num_threads), bool(use_bond_graph), frac_coords,
+            )
         (src_nodes, dst_nodes, markers, local_coords, global_ids,
          py_index_1, py_index_2, py_offsets, py_distances,
          line_src_nodes, line_dst_nodes, within_r_indices, line_markers,
@@ -146,7 +156,7 @@ class Distributed:
         else:
             line_markers = None
 
-        return cls(
+        obj = cls(
             src_nodes, dst_nodes, markers, local_coords, global_ids,
             py_index_1, py_index_2, py_offsets, py_distances,
             line_src_nodes, line_dst_nodes, within_r_indices, line_markers,
@@ -155,6 +165,8 @@ class Distributed:
             local_center_atom_indices_list, use_bond_graph, len(cart_coords),
             bde_global_edge_list=bde_global_edge_list,
         )
+        obj.csr_parts = csr_parts
+        return obj
 
     # -- size queries (reference dist.py:462-551) -------------------------
 

@@ -92,6 +92,20 @@ typedef struct {
                                       incl ghosts (build extension: lets a
                                       rank compute ghost bond geometry
                                       locally) */
+
+    /* Build extensions for the HIP kernels: local edges are emitted
+     * DST-SORTED (stable by global edge id within a dst), so the node
+     * scatter-add is a contiguous segmented reduction; every other
+     * scatter direction gets a permutation CSR (deterministic backward,
+     * no atomics).  Line edges are likewise emitted l_dst-sorted. */
+    const int64_t* row_ptr;            /* [n_nodes+1] CSR over dst_local   */
+    const int64_t* src_perm;           /* [n_edges]  edge ids sorted by src */
+    const int64_t* src_row_ptr;        /* [n_nodes+1]                      */
+    const int64_t* line_row_ptr;       /* [n_bonds+1] CSR over line_dst    */
+    const int64_t* line_src_perm;      /* [n_lines] line ids sorted by line_src */
+    const int64_t* line_src_row_ptr;   /* [n_bonds+1]                      */
+    const int64_t* center_perm;        /* [n_lines] line ids sorted by center atom */
+    const int64_t* center_row_ptr;     /* [n_nodes+1]                      */
 } dm_partition_view;
 
 int dm_graph_partition_view(const dm_graph* g, int32_t partition,
