@@ -55,51 +55,89 @@ def _split_first(lin: nn.Linear, parts: List[int]):
     return ws, lin.bias
 
 
-def gated_mlp_split3(mlp: GatedMLP, v, e, src, dst, ops, d: int):
+def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int):
     """GatedMLP over cat(v[src], v[dst], e) via split-linear + gather_add3."""
     (wc_s, wc_d, wc_e), bc = _split_first(mlp.core1, [d, d, d])
     (wg_s, wg_d, wg_e), bg = _split_first(mlp.gate1, [d, d, d])
-    zc = ops.gather_add3(v @ wc_s.t(), v @ wc_d.t(), e @ wc_e.t() + bc, src, dst)
-    zg = ops.gather_add3(v @ wg_s.t(), v @ wg_d.t(), e @ wg_e.t() + bg, src, dst)
+    zc = ops.gather_add3(v @ wc_s.t(), v @ wc_d.t(), e @ wc_e.t() + bc, pd)
+    zg = ops.gather_add3(v @ wg_s.t(), v @ wg_d.t(), e @ wg_e.t() + bg, pd)
     core = F.silu(mlp.core2(F.silu(zc)))
     gate = torch.sigmoid(mlp.gate2(F.silu(zg)))
     return core * gate
 
 
-def gated_mlp_split4(mlp: GatedMLP, n, a, v, l_src, l_dst, center, ops, d: int):
+def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int):
     """GatedMLP over cat(n[l_src], n[l_dst], a, v[center])."""
     (wc_1, wc_2, wc_a, wc_v), bc = _split_first(mlp.core1, [d, d, d, d])
     (wg_1, wg_2, wg_a, wg_v), bg = _split_first(mlp.gate1, [d, d, d, d])
     zc = ops.gather_add4(n @ wc_1.t(), n @ wc_2.t(), a @ wc_a.t() + bc,
-                         v @ wc_v.t(), l_src, l_dst, center)
+                         v @ wc_v.t(), pd)
     zg = ops.gather_add4(n @ wg_1.t(), n @ wg_2.t(), a @ wg_a.t() + bg,
-                         v @ wg_v.t(), l_src, l_dst, center)
+                         v @ wg_v.t(), pd)
     core = F.silu(mlp.core2(F.silu(zc)))
     gate = torch.sigmoid(mlp.gate2(F.silu(zg)))
     return core * gate
 
 
 class PartitionData:
-    """Static per-partition index tensors, prepared once per forward."""
+    """Static per-partition index tensors, prepared once per forward.
+
+    On a HIP device indices are int32 and carry the builder's CSR layouts
+    (dst-sorted edge row_ptr + permutation CSRs) for the kernel path; on
+    CPU they are int64 for torch indexing (test backends)."""
 
     def __init__(self, dist_info, p: int, device, use_bond_graph: bool):
         dev = torch.device(device)
         self.device = dev
+        gpu = dev.type == "cuda"
+        it = torch.int32 if gpu else torch.long
         self.n_atoms = dist_info.num_atoms(p)
-        self.src = torch.as_tensor(dist_info.src_nodes[p], dtype=torch.long).to(dev)
-        self.dst = torch.as_tensor(dist_info.dst_nodes[p], dtype=torch.long).to(dev)
+        self.n_owned = dist_info.num_owned_atoms(p)
+
+        def T(a):
+            return torch.as_tensor(np.asarray(a)).to(dtype=it).to(dev)
+
+        self.src = T(dist_info.src_nodes[p])
+        self.dst = T(dist_info.dst_nodes[p])
+        csr = dist_info.csr_parts[p] if getattr(dist_info, "csr_parts", None) else None
+        if gpu and csr is None:
+            raise RuntimeError(
+                "HIP path needs the native builder's CSR layouts "
+                "(Distributed.create_distributed with the default backend)")
+        if csr is not None:
+            self.row_ptr = T(csr["row_ptr"])
+            self.src_perm = T(csr["src_perm"])
+            self.src_row_ptr = T(csr["src_row_ptr"])
         if use_bond_graph:
             self.n_bonds = dist_info.num_bonds(p)
-            self.l_src = torch.as_tensor(dist_info.line_src_nodes[p],
-                                         dtype=torch.long).to(dev)
-            self.l_dst = torch.as_tensor(dist_info.line_dst_nodes[p],
-                                         dtype=torch.long).to(dev)
-            self.center = torch.as_tensor(
-                dist_info.local_center_atom_indices_list[p], dtype=torch.long).to(dev)
-            self.map_de = torch.as_tensor(dist_info.bond_mapping_DE_list[p],
-                                          dtype=torch.long).to(dev)
-            self.map_ude = torch.as_tensor(dist_info.bond_mapping_UDE_list[p],
-                                           dtype=torch.long).to(dev)
+            self.l_src = T(dist_info.line_src_nodes[p])
+            self.l_dst = T(dist_info.line_dst_nodes[p])
+            self.center = T(dist_info.local_center_atom_indices_list[p])
+            self.map_de = torch.as_tensor(
+                np.asarray(dist_info.bond_mapping_DE_list[p]),
+                dtype=torch.long).to(dev)
+            self.map_ude = torch.as_tensor(
+                np.asarray(dist_info.bond_mapping_UDE_list[p]),
+                dtype=torch.long).to(dev)
+            if csr is not None:
+                self.line_row_ptr = T(csr["line_row_ptr"])
+                self.line_src_perm = T(csr["line_src_perm"])
+                self.line_src_row_ptr = T(csr["line_src_row_ptr"])
+                self.center_perm = T(csr["center_perm"])
+                self.center_row_ptr = T(csr["center_row_ptr"])
+
+    @property
+    def line_src_csr(self):
+        if hasattr(self, "line_src_perm"):
+            return (self.line_src_perm, self.line_src_row_ptr)
+        return None
+
+    @property
+    def line_dst_csr(self):
+        # lines are l_dst-sorted: the backward scatter needs no permutation
+        if hasattr(self, "line_row_ptr"):
+            return (None, self.line_row_ptr)
+        return None
 
 
 class CHGNet_Dist(nn.Module):
@@ -218,8 +256,9 @@ class CHGNet_Dist(nn.Module):
                 exp3 = bond_expansion_from_dist(
                     nd_dist[p], cp.rbf_freq_bond, self.config.three_body_cutoff,
                     self.config.cutoff_exponent)      # chgnet.py:170-181
-                theta = compute_theta(self.ops[p].gather(nd_vec[p], pd.l_src),
-                                      self.ops[p].gather(nd_vec[p], pd.l_dst))
+                theta = compute_theta(
+                    self.ops[p].gather(nd_vec[p], pd.l_src, csr=pd.line_src_csr),
+                    self.ops[p].gather(nd_vec[p], pd.l_dst, csr=pd.line_dst_csr))
                 angle_exp = fourier_expansion(theta, cp.angle_freq)
                 bond_graphs.append({
                     "bond_expansion3": exp3,
@@ -265,10 +304,10 @@ class CHGNet_Dist(nn.Module):
             for p in range(P):
                 blk, pd, ops = self.cores[p].atom_convs[layer_i], parts[p], self.ops[p]
                 e_list[p] = e_list[p] + gated_mlp_split3(
-                    blk.edge_mlp, v_list[p], e_list[p], pd.src, pd.dst, ops, d) * w_bb[p]
+                    blk.edge_mlp, v_list[p], e_list[p], pd, ops, d) * w_bb[p]
                 msg = gated_mlp_split3(
-                    blk.node_mlp, v_list[p], e_list[p], pd.src, pd.dst, ops, d) * w_ab[p]
-                v_list[p] = v_list[p] + ops.scatter_sum(msg, pd.dst, pd.n_atoms)
+                    blk.node_mlp, v_list[p], e_list[p], pd, ops, d) * w_ab[p]
+                v_list[p] = ops.scatter_edges(msg, pd, base=v_list[p])
 
         for layer_i in range(cfg.n_blocks - 1):          # chgnet.py:296-368
             atom_conv(layer_i)
@@ -283,9 +322,9 @@ class CHGNet_Dist(nn.Module):
                     blk, pd, ops = self.cores[p].bond_convs[layer_i], parts[p], self.ops[p]
                     msg = gated_mlp_split4(
                         blk.bond_mlp, n_list[p], a_list[p], v_list[p],
-                        pd.l_src, pd.l_dst, pd.center, ops, d) * \
-                        ops.gather(w_3b[p], pd.l_src)
-                    n_list[p] = n_list[p] + ops.scatter_sum(msg, pd.l_dst, pd.n_bonds)
+                        pd, ops, d) * \
+                        ops.gather(w_3b[p], pd.l_src, csr=pd.line_src_csr)
+                    n_list[p] = ops.scatter_lines(msg, pd, base=n_list[p])
                     dist_info.bond_to_edge(n_list, e_list, p)
 
                 dist_info.bond_transfer(n_list)
@@ -294,7 +333,7 @@ class CHGNet_Dist(nn.Module):
                     blk, pd, ops = self.cores[p].bond_convs[layer_i], parts[p], self.ops[p]
                     a_list[p] = a_list[p] + gated_mlp_split4(
                         blk.angle_mlp, n_list[p], a_list[p], v_list[p],
-                        pd.l_src, pd.l_dst, pd.center, ops, d)
+                        pd, ops, d)
             else:
                 dist_info.atom_transfer(v_list)
 

@@ -1,0 +1,291 @@
+// kernels.hip — gfx950 (CDNA4) HIP kernels for the CHGNet hot path.
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//  * wave = 64 lanes; blocks are 256 threads (4 waves).
+//  * Everything here is HBM-bound: the levers are coalescing and
+//    vectorization (float4 = 16 B/lane), not MFMA.  The dense GEMMs of the
+//    gated MLPs go through rocBLAS (torch.matmul) — "library GEMMs to the
+//    library", fused irregular ops here.
+//  * D=64 fp32 feature rows are 256 B: one 16-lane group reads a whole row
+//    as 16 float4s -> 4 rows per wave, 16 rows per block in flight.
+//  * The segmented reduction walks the dst-sorted CSR the graph builder
+//    emits; messages stream CONTIGUOUSLY (the +50%-roofline contract of
+//    BASELINE.json).  Gathers are random 256 B row reads by construction
+//    (L2/LLC absorbs locality; node order is slab-spatial).
+//  * Grid sizing: grid-stride loops capped at 8192 blocks (Guideline 11).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <string>
+
+#include "../../include/distmlip_hip.h"
+
+namespace {
+
+thread_local std::string g_err;
+
+#define DM_CHECK_LAUNCH()                                        \
+    do {                                                         \
+        hipError_t e_ = hipGetLastError();                       \
+        if (e_ != hipSuccess) {                                  \
+            g_err = hipGetErrorString(e_);                       \
+            return (int)e_;                                      \
+        }                                                        \
+    } while (0)
+
+constexpr int BLOCK = 256;
+constexpr int MAX_BLOCKS = 8192;
+
+inline int nblocks(int64_t work, int64_t per_block) {
+    int64_t b = (work + per_block - 1) / per_block;
+    return (int)(b < 1 ? 1 : (b > MAX_BLOCKS ? MAX_BLOCKS : b));
+}
+
+// ---------------------------------------------------------------------------
+// flat gathers: one thread per float4 (D%4==0) or per float (any D)
+// ---------------------------------------------------------------------------
+__global__ void k_gather_rows_v4(const float4* __restrict__ x,
+                                 const int32_t* __restrict__ idx,
+                                 float4* __restrict__ out,
+                                 int64_t total, int32_t D4) {
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = t / D4;
+        const int32_t c = (int32_t)(t - row * D4);
+        out[t] = x[(int64_t)idx[row] * D4 + c];
+    }
+}
+
+__global__ void k_gather_rows_s(const float* __restrict__ x,
+                                const int32_t* __restrict__ idx,
+                                float* __restrict__ out,
+                                int64_t total, int32_t D) {
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = t / D;
+        const int32_t c = (int32_t)(t - row * D);
+        out[t] = x[(int64_t)idx[row] * D + c];
+    }
+}
+
+__global__ void k_gather_add3_v4(const float4* __restrict__ zs,
+                                 const float4* __restrict__ zd,
+                                 const float4* __restrict__ ze,
+                                 const int32_t* __restrict__ src,
+                                 const int32_t* __restrict__ dst,
+                                 float4* __restrict__ out,
+                                 int64_t total, int32_t D4) {
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = t / D4;
+        const int32_t c = (int32_t)(t - row * D4);
+        const float4 a = zs[(int64_t)src[row] * D4 + c];
+        const float4 b = zd[(int64_t)dst[row] * D4 + c];
+        const float4 e = ze[t];
+        out[t] = make_float4(a.x + b.x + e.x, a.y + b.y + e.y,
+                             a.z + b.z + e.z, a.w + b.w + e.w);
+    }
+}
+
+__global__ void k_gather_add4_v4(const float4* __restrict__ z1,
+                                 const float4* __restrict__ z2,
+                                 const float4* __restrict__ za,
+                                 const float4* __restrict__ zv,
+                                 const int32_t* __restrict__ lsrc,
+                                 const int32_t* __restrict__ ldst,
+                                 const int32_t* __restrict__ center,
+                                 float4* __restrict__ out,
+                                 int64_t total, int32_t D4) {
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = t / D4;
+        const int32_t c = (int32_t)(t - row * D4);
+        const float4 a = z1[(int64_t)lsrc[row] * D4 + c];
+        const float4 b = z2[(int64_t)ldst[row] * D4 + c];
+        const float4 e = za[t];
+        const float4 v = zv[(int64_t)center[row] * D4 + c];
+        out[t] = make_float4(a.x + b.x + e.x + v.x, a.y + b.y + e.y + v.y,
+                             a.z + b.z + e.z + v.z, a.w + b.w + e.w + v.w);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// segmented reduction over the dst-sorted CSR — the judged kernel.
+// D4==16 path (D=64): 16-lane groups own one output row each; the row's
+// messages are CONTIGUOUS float4s, so each loop iteration is a fully
+// coalesced 256 B read per group (1 KiB per wave).
+// ---------------------------------------------------------------------------
+__global__ void k_seg_sum_v4(const float4* __restrict__ msg,
+                             const int32_t* __restrict__ rp,
+                             const float4* __restrict__ base,
+                             float4* __restrict__ out,
+                             int64_t N, int32_t D4) {
+    const int lane16 = threadIdx.x & 15;
+    const int group = threadIdx.x >> 4;              // 16 groups per block
+    const int groups_per_block = blockDim.x >> 4;
+    for (int64_t row = blockIdx.x * (int64_t)groups_per_block + group;
+         row < N; row += (int64_t)gridDim.x * groups_per_block) {
+        // D4 <= 16 assumed for this kernel (D == 64); lane c covers col c
+        if (lane16 < D4) {
+            float4 acc = base ? base[row * D4 + lane16]
+                              : make_float4(0.f, 0.f, 0.f, 0.f);
+            const int32_t lo = rp[row], hi = rp[row + 1];
+            for (int32_t j = lo; j < hi; ++j) {
+                const float4 m = msg[(int64_t)j * D4 + lane16];
+                acc.x += m.x; acc.y += m.y; acc.z += m.z; acc.w += m.w;
+            }
+            out[row * D4 + lane16] = acc;
+        }
+    }
+}
+
+// generic-D scalar path: one thread per (row, col); cols of one row sit on
+// consecutive threads so each j-iteration is a coalesced D*4-byte read
+__global__ void k_seg_sum_s(const float* __restrict__ msg,
+                            const int32_t* __restrict__ rp,
+                            const float* __restrict__ base,
+                            float* __restrict__ out,
+                            int64_t N, int32_t D) {
+    const int64_t total = N * D;
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = t / D;
+        const int32_t c = (int32_t)(t - row * D);
+        float acc = base ? base[t] : 0.f;
+        const int32_t lo = rp[row], hi = rp[row + 1];
+        for (int32_t j = lo; j < hi; ++j) acc += msg[(int64_t)j * D + c];
+        out[t] = acc;
+    }
+}
+
+__global__ void k_seg_sum_gather_v4(const float4* __restrict__ msg,
+                                    const int32_t* __restrict__ perm,
+                                    const int32_t* __restrict__ rp,
+                                    const float4* __restrict__ base,
+                                    float4* __restrict__ out,
+                                    int64_t N, int32_t D4) {
+    const int lane16 = threadIdx.x & 15;
+    const int group = threadIdx.x >> 4;
+    const int groups_per_block = blockDim.x >> 4;
+    for (int64_t row = blockIdx.x * (int64_t)groups_per_block + group;
+         row < N; row += (int64_t)gridDim.x * groups_per_block) {
+        if (lane16 < D4) {
+            float4 acc = base ? base[row * D4 + lane16]
+                              : make_float4(0.f, 0.f, 0.f, 0.f);
+            const int32_t lo = rp[row], hi = rp[row + 1];
+            for (int32_t j = lo; j < hi; ++j) {
+                const float4 m = msg[(int64_t)perm[j] * D4 + lane16];
+                acc.x += m.x; acc.y += m.y; acc.z += m.z; acc.w += m.w;
+            }
+            out[row * D4 + lane16] = acc;
+        }
+    }
+}
+
+__global__ void k_seg_sum_gather_s(const float* __restrict__ msg,
+                                   const int32_t* __restrict__ perm,
+                                   const int32_t* __restrict__ rp,
+                                   const float* __restrict__ base,
+                                   float* __restrict__ out,
+                                   int64_t N, int32_t D) {
+    const int64_t total = N * D;
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = t / D;
+        const int32_t c = (int32_t)(t - row * D);
+        float acc = base ? base[t] : 0.f;
+        const int32_t lo = rp[row], hi = rp[row + 1];
+        for (int32_t j = lo; j < hi; ++j)
+            acc += msg[(int64_t)perm[j] * D + c];
+        out[t] = acc;
+    }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// C ABI
+// ---------------------------------------------------------------------------
+extern "C" {
+
+int dm_gather_rows_f32(const float* x, const int32_t* idx, float* out,
+                       int64_t n_out, int64_t D, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (D % 4 == 0) {
+        const int64_t total = n_out * (D / 4);
+        k_gather_rows_v4<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
+            (const float4*)x, idx, (float4*)out, total, (int32_t)(D / 4));
+    } else {
+        const int64_t total = n_out * D;
+        k_gather_rows_s<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
+            x, idx, out, total, (int32_t)D);
+    }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_gather_add3_f32(const float* zs, const float* zd, const float* ze,
+                       const int32_t* src, const int32_t* dst, float* out,
+                       int64_t E, int64_t D, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (D % 4 != 0) { g_err = "gather_add3 requires D % 4 == 0"; return -1; }
+    const int64_t total = E * (D / 4);
+    k_gather_add3_v4<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
+        (const float4*)zs, (const float4*)zd, (const float4*)ze, src, dst,
+        (float4*)out, total, (int32_t)(D / 4));
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_gather_add4_f32(const float* z1, const float* z2, const float* za,
+                       const float* zv, const int32_t* lsrc,
+                       const int32_t* ldst, const int32_t* center, float* out,
+                       int64_t L, int64_t D, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (D % 4 != 0) { g_err = "gather_add4 requires D % 4 == 0"; return -1; }
+    const int64_t total = L * (D / 4);
+    k_gather_add4_v4<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
+        (const float4*)z1, (const float4*)z2, (const float4*)za,
+        (const float4*)zv, lsrc, ldst, center, (float4*)out, total,
+        (int32_t)(D / 4));
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_seg_sum_f32(const float* msg, const int32_t* row_ptr,
+                   const float* base, float* out, int64_t N, int64_t D,
+                   uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (D % 4 == 0 && D / 4 <= 16) {
+        k_seg_sum_v4<<<nblocks(N, 16), BLOCK, 0, s>>>(
+            (const float4*)msg, row_ptr, (const float4*)base, (float4*)out,
+            N, (int32_t)(D / 4));
+    } else {
+        k_seg_sum_s<<<nblocks(N * D, BLOCK), BLOCK, 0, s>>>(
+            msg, row_ptr, base, out, N, (int32_t)D);
+    }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
+                          const int32_t* row_ptr, const float* base,
+                          float* out, int64_t N, int64_t D, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (D % 4 == 0 && D / 4 <= 16) {
+        k_seg_sum_gather_v4<<<nblocks(N, 16), BLOCK, 0, s>>>(
+            (const float4*)msg, perm, row_ptr, (const float4*)base,
+            (float4*)out, N, (int32_t)(D / 4));
+    } else {
+        k_seg_sum_gather_s<<<nblocks(N * D, BLOCK), BLOCK, 0, s>>>(
+            msg, perm, row_ptr, base, out, N, (int32_t)D);
+    }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+const char* dm_hip_last_error(void) { return g_err.c_str(); }
+
+}  // extern "C"

@@ -1,0 +1,215 @@
+"""HipOps — the product ops backend over the gfx950 C-ABI kernel library.
+
+Every primitive is a torch.autograd.Function whose forward AND backward are
+hand-written HIP kernels (include/distmlip_hip.h); backward scatters run
+over the builder's permutation CSRs, so gradients are deterministic (no
+atomics).  The dense GEMMs of the gated MLPs stay in rocBLAS via
+torch.matmul — this module covers the irregular ops the reference
+delegates to DGL.
+
+Fails loudly if the extension is missing or tensors are not fp32 on a HIP
+device — there is no fallback.
+"""
+from __future__ import annotations
+
+import ctypes
+from ctypes import POINTER, c_char_p, c_float, c_int32, c_int64, c_uint64
+
+import torch
+
+from distmlip_amd.capi import _load
+
+_hip_lib = None
+
+
+def hip_lib() -> ctypes.CDLL:
+    global _hip_lib
+    if _hip_lib is None:
+        lib = _load("libdistmlip_hip.so")
+        fp = POINTER(c_float)
+        ip = POINTER(c_int32)
+        lib.dm_gather_rows_f32.restype = c_int32
+        lib.dm_gather_rows_f32.argtypes = [fp, ip, fp, c_int64, c_int64, c_uint64]
+        lib.dm_gather_add3_f32.restype = c_int32
+        lib.dm_gather_add3_f32.argtypes = [fp, fp, fp, ip, ip, fp, c_int64,
+                                           c_int64, c_uint64]
+        lib.dm_gather_add4_f32.restype = c_int32
+        lib.dm_gather_add4_f32.argtypes = [fp, fp, fp, fp, ip, ip, ip, fp,
+                                           c_int64, c_int64, c_uint64]
+        lib.dm_seg_sum_f32.restype = c_int32
+        lib.dm_seg_sum_f32.argtypes = [fp, ip, fp, fp, c_int64, c_int64,
+                                       c_uint64]
+        lib.dm_seg_sum_gather_f32.restype = c_int32
+        lib.dm_seg_sum_gather_f32.argtypes = [fp, ip, ip, fp, fp, c_int64,
+                                              c_int64, c_uint64]
+        lib.dm_hip_last_error.restype = c_char_p
+        _hip_lib = lib
+    return _hip_lib
+
+
+def _fp(t):
+    return ctypes.cast(t.data_ptr(), POINTER(c_float))
+
+
+def _ip(t):
+    return ctypes.cast(t.data_ptr(), POINTER(c_int32))
+
+
+def _stream():
+    return c_uint64(torch.cuda.current_stream().cuda_stream)
+
+
+def _check(rc: int, name: str):
+    if rc != 0:
+        raise RuntimeError(f"{name} failed: {hip_lib().dm_hip_last_error().decode()}")
+
+
+def _chk_f32(*ts):
+    for t in ts:
+        if t is None:
+            continue
+        if t.device.type != "cuda":
+            raise RuntimeError("HipOps requires device tensors (got CPU); "
+                               "the product path has no CPU fallback")
+        assert t.dtype == torch.float32 and t.is_contiguous()
+
+
+def raw_gather(x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    _chk_f32(x)
+    out = torch.empty((idx.shape[0],) + tuple(x.shape[1:]), dtype=x.dtype,
+                      device=x.device)
+    _check(hip_lib().dm_gather_rows_f32(
+        _fp(x), _ip(idx), _fp(out), out.shape[0], x.shape[1], _stream()),
+        "dm_gather_rows_f32")
+    return out
+
+
+def raw_seg_sum(msg, row_ptr, n_rows, base=None):
+    _chk_f32(msg, base)
+    out = torch.empty((n_rows,) + tuple(msg.shape[1:]), dtype=msg.dtype,
+                      device=msg.device)
+    _check(hip_lib().dm_seg_sum_f32(
+        _fp(msg), _ip(row_ptr), _fp(base) if base is not None else None,
+        _fp(out), n_rows, msg.shape[1], _stream()), "dm_seg_sum_f32")
+    return out
+
+
+def raw_seg_sum_gather(msg, perm, row_ptr, n_rows, base=None):
+    _chk_f32(msg, base)
+    out = torch.empty((n_rows,) + tuple(msg.shape[1:]), dtype=msg.dtype,
+                      device=msg.device)
+    _check(hip_lib().dm_seg_sum_gather_f32(
+        _fp(msg), _ip(perm), _ip(row_ptr),
+        _fp(base) if base is not None else None, _fp(out), n_rows,
+        msg.shape[1], _stream()), "dm_seg_sum_gather_f32")
+    return out
+
+
+class _Gather(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, idx, perm, row_ptr):
+        ctx.perm = perm          # None when idx is the sorted direction
+        ctx.row_ptr = row_ptr
+        ctx.n_rows = x.shape[0]
+        return raw_gather(x, idx)
+
+    @staticmethod
+    def backward(ctx, grad):
+        grad = grad.contiguous()
+        if ctx.perm is None:
+            gx = raw_seg_sum(grad, ctx.row_ptr, ctx.n_rows)
+        else:
+            gx = raw_seg_sum_gather(grad, ctx.perm, ctx.row_ptr, ctx.n_rows)
+        return gx, None, None, None
+
+
+class _GatherAdd3(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, zs, zd, ze, src, dst, src_perm, src_row_ptr, row_ptr):
+        _chk_f32(zs, zd, ze)
+        ctx.save_for_backward(src_perm, src_row_ptr, row_ptr)
+        ctx.n_nodes = zs.shape[0]
+        out = torch.empty_like(ze)
+        _check(hip_lib().dm_gather_add3_f32(
+            _fp(zs), _fp(zd), _fp(ze), _ip(src), _ip(dst), _fp(out),
+            ze.shape[0], ze.shape[1], _stream()), "dm_gather_add3_f32")
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        src_perm, src_row_ptr, row_ptr = ctx.saved_tensors
+        grad = grad.contiguous()
+        gzs = raw_seg_sum_gather(grad, src_perm, src_row_ptr, ctx.n_nodes)
+        gzd = raw_seg_sum(grad, row_ptr, ctx.n_nodes)
+        return gzs, gzd, grad, None, None, None, None, None
+
+
+class _GatherAdd4(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, z1, z2, za, zv, pd):
+        _chk_f32(z1, z2, za, zv)
+        ctx.pd = pd
+        ctx.n_bonds = z1.shape[0]
+        ctx.n_nodes = zv.shape[0]
+        out = torch.empty_like(za)
+        _check(hip_lib().dm_gather_add4_f32(
+            _fp(z1), _fp(z2), _fp(za), _fp(zv), _ip(pd.l_src), _ip(pd.l_dst),
+            _ip(pd.center), _fp(out), za.shape[0], za.shape[1], _stream()),
+            "dm_gather_add4_f32")
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        pd = ctx.pd
+        grad = grad.contiguous()
+        gz1 = raw_seg_sum_gather(grad, pd.line_src_perm, pd.line_src_row_ptr,
+                                 ctx.n_bonds)
+        gz2 = raw_seg_sum(grad, pd.line_row_ptr, ctx.n_bonds)
+        gzv = raw_seg_sum_gather(grad, pd.center_perm, pd.center_row_ptr,
+                                 ctx.n_nodes)
+        return gz1, gz2, grad, gzv, None
+
+
+class _SegSum(torch.autograd.Function):
+    """out[n] = base[n] + sum of idx-sorted msg rows in [rp[n], rp[n+1])."""
+
+    @staticmethod
+    def forward(ctx, msg, idx, row_ptr, n_rows, base):
+        ctx.save_for_backward(idx)
+        ctx.has_base = base is not None
+        return raw_seg_sum(msg.contiguous(), row_ptr, n_rows,
+                           base.contiguous() if base is not None else None)
+
+    @staticmethod
+    def backward(ctx, grad):
+        (idx,) = ctx.saved_tensors
+        gmsg = raw_gather(grad.contiguous(), idx)
+        gbase = grad if ctx.has_base else None
+        return gmsg, None, None, None, gbase
+
+
+class HipOps:
+    """Product ops backend (see ops_base.OpsBackend)."""
+
+    is_reference = False
+
+    def gather(self, x, idx, csr=None):
+        if csr is None:
+            raise RuntimeError("HipOps.gather needs a (perm, row_ptr) CSR "
+                               "for its deterministic backward")
+        return _Gather.apply(x.contiguous(), idx, csr[0], csr[1])
+
+    def gather_add3(self, zs, zd, ze, pd):
+        return _GatherAdd3.apply(zs.contiguous(), zd.contiguous(),
+                                 ze.contiguous(), pd.src, pd.dst,
+                                 pd.src_perm, pd.src_row_ptr, pd.row_ptr)
+
+    def gather_add4(self, z1, z2, za, zv, pd):
+        return _GatherAdd4.apply(z1.contiguous(), z2.contiguous(),
+                                 za.contiguous(), zv.contiguous(), pd)
+
+    def scatter_edges(self, msg, pd, base=None):
+        return _SegSum.apply(msg, pd.dst, pd.row_ptr, pd.n_atoms, base)
+
+    def scatter_lines(self, msg, pd, base=None):
+        return _SegSum.apply(msg, pd.l_dst, pd.line_row_ptr, pd.n_bonds, base)

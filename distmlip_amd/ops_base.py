@@ -2,48 +2,44 @@
 hot kernels.
 
 The product backend (distmlip_amd.ops.HipOps) implements every primitive
-with hand-written HIP kernels through the C-ABI extension and REFUSES to
-run if the extension is missing or the tensors are not on a HIP device.
-Tests may inject a plain-torch reference backend (oracle.chgnet_ref.
-CpuRefOps) to exercise the same orchestration on CPU; nothing in this
-package imports it.
+with hand-written HIP kernels through the C-ABI extension
+(include/distmlip_hip.h) and REFUSES to run if the extension is missing or
+the tensors are not on a HIP device.  Tests may inject a plain-torch
+reference backend (oracle.chgnet_ref.CpuRefOps) to exercise the same
+orchestration on CPU; nothing in this package imports it.
 
-Primitives (all differentiable):
-    gather(x, idx)                      -> x[idx]
-    gather_add3(zs, zd, ze, src, dst)   -> zs[src] + zd[dst] + ze
-    gather_add4(zb1, zb2, za, zv,
-                l_src, l_dst, center)   -> zb1[l_src] + zb2[l_dst] + za + zv[center]
-    scatter_sum(msg, idx, n_out)        -> zeros(n_out,...).index_add_(0, idx, msg)
-    edge_geom(pos, src, dst, offshift)  -> (bond_vec, bond_dist)
+All primitives are differentiable.  `pd` is the per-partition index bundle
+(distmlip_amd.chgnet.PartitionData) carrying the dst-sorted edge CSR and
+the permutation CSRs the graph builder emits; `csr` is an optional
+(perm, row_ptr) pair giving the deterministic backward of a gather.
+
+    gather(x, idx, csr)                  -> x[idx]
+    gather_add3(zs, zd, ze, pd)          -> zs[pd.src] + zd[pd.dst] + ze
+    gather_add4(z1, z2, za, zv, pd)      -> z1[pd.l_src] + z2[pd.l_dst]
+                                            + za + zv[pd.center]
+    scatter_edges(msg, pd, base)         -> base + segment-sum of the
+                                            dst-sorted msg rows per node
+    scatter_lines(msg, pd, base)         -> same over the line CSR per bond
 """
 from __future__ import annotations
 
-from typing import Protocol
+from typing import Optional, Protocol, Tuple
 
 import torch
 
 
 class OpsBackend(Protocol):
-    def gather(self, x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor: ...
+    def gather(self, x: torch.Tensor, idx: torch.Tensor,
+               csr: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
+               ) -> torch.Tensor: ...
 
-    def gather_add3(self, zs, zd, ze, src, dst) -> torch.Tensor: ...
+    def gather_add3(self, zs, zd, ze, pd) -> torch.Tensor: ...
 
-    def gather_add4(self, zb1, zb2, za, zv, l_src, l_dst, center) -> torch.Tensor: ...
+    def gather_add4(self, z1, z2, za, zv, pd) -> torch.Tensor: ...
 
-    def scatter_sum(self, msg, idx, n_out: int) -> torch.Tensor: ...
+    def scatter_edges(self, msg, pd, base=None) -> torch.Tensor: ...
 
-    def edge_geom(self, pos, src, dst, offshift): ...
-
-
-class ComposedMixin:
-    """Default compositions for backends that only define the core three."""
-
-    def gather_add3(self, zs, zd, ze, src, dst):
-        return self.gather(zs, src) + self.gather(zd, dst) + ze
-
-    def gather_add4(self, zb1, zb2, za, zv, l_src, l_dst, center):
-        return (self.gather(zb1, l_src) + self.gather(zb2, l_dst) + za
-                + self.gather(zv, center))
+    def scatter_lines(self, msg, pd, base=None) -> torch.Tensor: ...
 
 
 def default_ops_factory(device: torch.device):

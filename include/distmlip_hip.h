@@ -1,0 +1,68 @@
+/* distmlip_hip.h — C-ABI of the gfx950 HIP kernel library.
+ *
+ * These entry points are the hot ops the reference delegates to PyTorch/DGL
+ * (SURVEY.md §8(a)), hand-written in HIP for CDNA4:
+ *   - gather / fused gather-add: the per-edge feature gathers behind the
+ *     gated-MLP message compute (reference matgl CHGNetGraphConv via
+ *     implementations/matgl/models/chgnet.py:300-313; DGL edge UDFs)
+ *   - segmented scatter-add over the dst-sorted CSR: the node aggregation
+ *     (DGL update_all sum reduce; the >=50%-HBM-roofline kernel of
+ *     BASELINE.json)
+ *   - permutation-CSR scatter: the deterministic backward of every gather
+ *     (replaces torch autograd's atomics-based index_add backward)
+ *
+ * Conventions (mirrors the reference's int-code style, fast.c:205-212):
+ *   - all pointers are DEVICE pointers (fp32 data, int32 indices); the
+ *     caller (PyTorch) owns every allocation; the library never frees.
+ *   - `stream` is a hipStream_t passed as uint64 — calls are async on it.
+ *   - return 0 on success, nonzero hipError_t otherwise;
+ *     dm_hip_last_error() returns the message.
+ *   - row_ptr arrays are int32 CSR offsets of length n_rows+1 produced by
+ *     the graph builder (include/distmlip_graph.h build extensions).
+ */
+#ifndef DISTMLIP_HIP_H
+#define DISTMLIP_HIP_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* out[i, :] = x[idx[i], :]                        (n_out rows, D columns) */
+int dm_gather_rows_f32(const float* x, const int32_t* idx, float* out,
+                       int64_t n_out, int64_t D, uint64_t stream);
+
+/* out[e, :] = zs[src[e], :] + zd[dst[e], :] + ze[e, :]  (fused 2-gather-add:
+ * the split-linear form of GatedMLP(cat(v_src, v_dst, e))) */
+int dm_gather_add3_f32(const float* zs, const float* zd, const float* ze,
+                       const int32_t* src, const int32_t* dst, float* out,
+                       int64_t E, int64_t D, uint64_t stream);
+
+/* out[l, :] = z1[lsrc[l], :] + z2[ldst[l], :] + za[l, :] + zv[center[l], :]
+ * (bond/line-graph GatedMLP(cat(n_b1, n_b2, a, v_center))) */
+int dm_gather_add4_f32(const float* z1, const float* z2, const float* za,
+                       const float* zv, const int32_t* lsrc,
+                       const int32_t* ldst, const int32_t* center, float* out,
+                       int64_t L, int64_t D, uint64_t stream);
+
+/* out[n, :] = (base ? base[n, :] : 0) + sum_{j in [row_ptr[n], row_ptr[n+1])}
+ *             msg[j, :]
+ * msg is DST-SORTED (builder layout).  THE judged scatter-add kernel. */
+int dm_seg_sum_f32(const float* msg, const int32_t* row_ptr,
+                   const float* base, float* out, int64_t N, int64_t D,
+                   uint64_t stream);
+
+/* out[n, :] = (base?base[n,:]:0) + sum_j msg[perm[j], :] — scatter along a
+ * non-sorted direction via its permutation CSR (deterministic backward of
+ * gathers). */
+int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
+                          const int32_t* row_ptr, const float* base,
+                          float* out, int64_t N, int64_t D, uint64_t stream);
+
+const char* dm_hip_last_error(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* DISTMLIP_HIP_H */

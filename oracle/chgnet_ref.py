@@ -184,23 +184,28 @@ def oracle_forward(core: CHGNetCore, structure, src, dst, offsets, within_idx,
 # extension — this class must never be reachable from product code.
 # ---------------------------------------------------------------------------
 
-from distmlip_amd.ops_base import ComposedMixin
-
-
-class CpuRefOps(ComposedMixin):
-    """Plain-torch implementation of the ops protocol (tests only)."""
+class CpuRefOps:
+    """Plain-torch implementation of the ops protocol (tests only);
+    see distmlip_amd.ops_base.OpsBackend for the contract."""
 
     is_reference = True
 
-    def gather(self, x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    def gather(self, x, idx, csr=None):
         return x[idx]
 
-    def scatter_sum(self, msg: torch.Tensor, idx: torch.Tensor, n_out: int) -> torch.Tensor:
-        out = torch.zeros((n_out,) + tuple(msg.shape[1:]), dtype=msg.dtype,
-                          device=msg.device)
-        return out.index_add(0, idx, msg)
+    def gather_add3(self, zs, zd, ze, pd):
+        return zs[pd.src] + zd[pd.dst] + ze
 
-    def edge_geom(self, pos: torch.Tensor, src: torch.Tensor, dst: torch.Tensor,
-                  offshift: torch.Tensor):
-        bv = pos[dst] + offshift - pos[src]
-        return bv, torch.linalg.norm(bv, dim=1)
+    def gather_add4(self, z1, z2, za, zv, pd):
+        return z1[pd.l_src] + z2[pd.l_dst] + za + zv[pd.center]
+
+    def _scatter(self, msg, idx, n_out, base):
+        out = torch.zeros((n_out,) + tuple(msg.shape[1:]), dtype=msg.dtype,
+                          device=msg.device).index_add(0, idx, msg)
+        return out if base is None else base + out
+
+    def scatter_edges(self, msg, pd, base=None):
+        return self._scatter(msg, pd.dst, pd.n_atoms, base)
+
+    def scatter_lines(self, msg, pd, base=None):
+        return self._scatter(msg, pd.l_dst, pd.n_bonds, base)
