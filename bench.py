@@ -45,6 +45,8 @@ def parse_args():
                     help="graph-builder threads (0 = cpu_count/world)")
     ap.add_argument("--no-bond-graph", action="store_true")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--breakdown", action="store_true",
+                    help="print per-stage timings (graph/H2D+fwd/bwd)")
     return ap.parse_args()
 
 
@@ -111,7 +113,7 @@ def cpu_baseline_leg(workload_name, threads):
     from distmlip_amd.structures import diamond_si, workload
     from oracle.chgnet_ref import oracle_forward
 
-    s = diamond_si(8, jitter=0.1, seed=0)        # 4096-atom sample
+    s = diamond_si(6, jitter=0.1, seed=0)        # 1,728-atom bounded sample
     torch.set_num_threads(threads)
     t0 = time.time()
     d = Distributed.create_distributed(
@@ -127,7 +129,7 @@ def cpu_baseline_leg(workload_name, threads):
         "cores": threads,
         "kind": "port",
         "sample": f"diamond-Si {s.num_atoms} atoms, 1 full E+F forward "
-                  f"(graph build + oracle model), {dt:.2f}s",
+                  f"(graph build + oracle CPU restatement), {dt:.2f}s",
     }
 
 
@@ -167,6 +169,43 @@ def main():
         def step():
             E, F, _, _ = pot.forward(s)
             return E, F
+
+        if args.breakdown:
+            stages = {}
+
+            def step():  # noqa: F811
+                from distmlip_amd.dist import Distributed
+
+                def mark(name, t0):
+                    torch.cuda.synchronize()
+                    stages.setdefault(name, []).append(time.time() - t0)
+                    return time.time()
+
+                t = time.time()
+                lattice_matrix = np.asarray(s.lattice, dtype=float)
+                d = Distributed.create_distributed(
+                    s.cart_coords, s.frac_coords, lattice_matrix, 1, s.pbc,
+                    6.0, 3.0, use_bond_graph=use_bg, num_threads=threads)
+                t = mark("graph_cpu", t)
+                model.set_local_species(d, s.species)
+                out = model.potential_forward_dist(
+                    d, s, lattice_matrix, False, True, False, None)
+                t = mark("h2d+forward", t)
+                node_types, positions, strain, (E, site) = out
+                E = model.core.data_std * E + model.core.data_mean
+                E = E + model.cores[0].element_refs[node_types].sum()
+                torch.autograd.backward(E)
+                F = -positions.grad
+                t = mark("backward", t)
+                return E, F
+
+            import atexit
+
+            def report():
+                for k, v in stages.items():
+                    print(f"# stage {k}: {1e3*sum(v)/len(v):.1f} ms avg over "
+                          f"{len(v)} calls", file=sys.stderr)
+            atexit.register(report)
 
     timer = SegSumTimer().wrap()
 

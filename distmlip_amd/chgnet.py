@@ -327,13 +327,16 @@ class CHGNet_Dist(nn.Module):
                     n_list[p] = ops.scatter_lines(msg, pd, base=n_list[p])
                     dist_info.bond_to_edge(n_list, e_list, p)
 
-                dist_info.bond_transfer(n_list)
-
-                for p in range(P):                       # angle pass chgnet.py:353-368
-                    blk, pd, ops = self.cores[p].bond_convs[layer_i], parts[p], self.ops[p]
-                    a_list[p] = a_list[p] + gated_mlp_split4(
-                        blk.angle_mlp, n_list[p], a_list[p], v_list[p],
-                        pd, ops, d)
+                if layer_i < cfg.n_blocks - 2:
+                    # the last bond block's angle update is dead compute
+                    # (the reference still runs it, chgnet.py:353-368;
+                    # numerics are identical without it) — skip
+                    dist_info.bond_transfer(n_list)
+                    for p in range(P):                   # angle pass chgnet.py:353-368
+                        blk, pd, ops = self.cores[p].bond_convs[layer_i], parts[p], self.ops[p]
+                        a_list[p] = a_list[p] + gated_mlp_split4(
+                            blk.angle_mlp, n_list[p], a_list[p], v_list[p],
+                            pd, ops, d)
             else:
                 dist_info.atom_transfer(v_list)
 

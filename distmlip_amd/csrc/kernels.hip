@@ -129,14 +129,20 @@ __global__ void k_seg_sum_v4(const float4* __restrict__ msg,
          row < N; row += (int64_t)gridDim.x * groups_per_block) {
         // D4 <= 16 assumed for this kernel (D == 64); lane c covers col c
         if (lane16 < D4) {
-            float4 acc = base ? base[row * D4 + lane16]
-                              : make_float4(0.f, 0.f, 0.f, 0.f);
+            // fp64 accumulate (register-only; kernel is HBM-bound so the
+            // extra VALU is free) — tightens force parity vs the oracle
+            double ax = 0, ay = 0, az = 0, aw = 0;
+            if (base) {
+                const float4 b = base[row * D4 + lane16];
+                ax = b.x; ay = b.y; az = b.z; aw = b.w;
+            }
             const int32_t lo = rp[row], hi = rp[row + 1];
             for (int32_t j = lo; j < hi; ++j) {
                 const float4 m = msg[(int64_t)j * D4 + lane16];
-                acc.x += m.x; acc.y += m.y; acc.z += m.z; acc.w += m.w;
+                ax += m.x; ay += m.y; az += m.z; aw += m.w;
             }
-            out[row * D4 + lane16] = acc;
+            out[row * D4 + lane16] = make_float4((float)ax, (float)ay,
+                                                 (float)az, (float)aw);
         }
     }
 }
@@ -153,10 +159,10 @@ __global__ void k_seg_sum_s(const float* __restrict__ msg,
          t < total; t += (int64_t)gridDim.x * blockDim.x) {
         const int64_t row = t / D;
         const int32_t c = (int32_t)(t - row * D);
-        float acc = base ? base[t] : 0.f;
+        double acc = base ? (double)base[t] : 0.0;
         const int32_t lo = rp[row], hi = rp[row + 1];
         for (int32_t j = lo; j < hi; ++j) acc += msg[(int64_t)j * D + c];
-        out[t] = acc;
+        out[t] = (float)acc;
     }
 }
 
@@ -172,14 +178,18 @@ __global__ void k_seg_sum_gather_v4(const float4* __restrict__ msg,
     for (int64_t row = blockIdx.x * (int64_t)groups_per_block + group;
          row < N; row += (int64_t)gridDim.x * groups_per_block) {
         if (lane16 < D4) {
-            float4 acc = base ? base[row * D4 + lane16]
-                              : make_float4(0.f, 0.f, 0.f, 0.f);
+            double ax = 0, ay = 0, az = 0, aw = 0;
+            if (base) {
+                const float4 b = base[row * D4 + lane16];
+                ax = b.x; ay = b.y; az = b.z; aw = b.w;
+            }
             const int32_t lo = rp[row], hi = rp[row + 1];
             for (int32_t j = lo; j < hi; ++j) {
                 const float4 m = msg[(int64_t)perm[j] * D4 + lane16];
-                acc.x += m.x; acc.y += m.y; acc.z += m.z; acc.w += m.w;
+                ax += m.x; ay += m.y; az += m.z; aw += m.w;
             }
-            out[row * D4 + lane16] = acc;
+            out[row * D4 + lane16] = make_float4((float)ax, (float)ay,
+                                                 (float)az, (float)aw);
         }
     }
 }
@@ -195,11 +205,11 @@ __global__ void k_seg_sum_gather_s(const float* __restrict__ msg,
          t < total; t += (int64_t)gridDim.x * blockDim.x) {
         const int64_t row = t / D;
         const int32_t c = (int32_t)(t - row * D);
-        float acc = base ? base[t] : 0.f;
+        double acc = base ? (double)base[t] : 0.0;
         const int32_t lo = rp[row], hi = rp[row + 1];
         for (int32_t j = lo; j < hi; ++j)
             acc += msg[(int64_t)perm[j] * D + c];
-        out[t] = acc;
+        out[t] = (float)acc;
     }
 }
 

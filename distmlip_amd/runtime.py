@@ -1,0 +1,324 @@
+"""SPMD runtime — one process per GPU over torch.distributed (RCCL/xGMI).
+
+The MI355X-native replacement for the reference's single-process
+multi-GPU loop (chgnet.py:296-368 + the synchronous p2p slice copies of
+dist.py:356): each rank owns one slab partition, computes its partition's
+kernels, and exchanges border-node features with its slab neighbors via
+point-to-point isend/irecv (RCCL over xGMI; slab topology means <= 2 peers
+per rank).  The ONLY collective on the data path is the final scalar
+energy all-reduce (SURVEY.md §2 "MI355X-native equivalent").
+
+`HaloExchange` is an autograd.Function: forward replaces each ghost
+("from") slice with the owner's fresh "to" slice; backward routes the
+ghost-slice gradients back to the owners and ADDS them into the "to"
+slices' gradients (the distributed transpose of the copy).  This is what
+makes `torch.autograd.backward(E_local)` on every rank produce exactly the
+reference's forces (pes.py:121-124) without a central GPU.
+
+Divergences from the reference flow (documented in DESIGN.md):
+  * positions and bond geometry are computed PER-RANK from the replicated
+    structure (removes the GPU0-centralized bond_vec serialization point,
+    chgnet.py:96-100); ghost BOND geometry still arrives by one
+    bond-halo at forward start, exactly like chgnet.py:157-164.
+  * forces are assembled by one reverse halo-add of position gradients;
+    each rank ends with exact forces for its owned atoms.
+"""
+from __future__ import annotations
+
+from copy import deepcopy
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from distmlip_amd.chgnet import (
+    CHGNet_Dist,
+    PartitionData,
+    gated_mlp_split3,
+    gated_mlp_split4,
+)
+from distmlip_amd.dist import Distributed
+from distmlip_amd.model import (
+    CHGNetCore,
+    bond_expansion_from_dist,
+    compute_theta,
+    fourier_expansion,
+)
+from distmlip_amd.ops_base import default_ops_factory
+
+
+def halo_plan(markers, rank: int, P: int) -> List[Tuple[int, int, int, int, int]]:
+    """[(peer, send_start, send_end, recv_start, recv_end)] from the marker
+    contract (dist.py:44-51): send my 'to peer' slice, receive into my
+    'from peer' slice."""
+    plan = []
+    for q in range(P):
+        if q == rank:
+            continue
+        ss, se = int(markers[1 + q]), int(markers[1 + q + 1])
+        rs, re = int(markers[1 + P + q]), int(markers[1 + P + q + 1])
+        if ss != se or rs != re:
+            plan.append((q, ss, se, rs, re))
+    return plan
+
+
+def _exchange(feat: torch.Tensor, plan, reverse: bool = False) -> dict:
+    """Post all irecvs then isends, wait; returns {peer: recv_buffer}.
+    reverse=True swaps roles (send my recv-slices, receive for my
+    send-slices) — the backward direction."""
+    recvs, reqs = {}, []
+    for (q, ss, se, rs, re) in plan:
+        a, b = (rs, re) if reverse else (ss, se)
+        c, d = (ss, se) if reverse else (rs, re)
+        if d > c:
+            buf = torch.empty((d - c,) + tuple(feat.shape[1:]),
+                              dtype=feat.dtype, device=feat.device)
+            recvs[q] = buf
+            reqs.append(dist.irecv(buf, src=q))
+    sends = []
+    for (q, ss, se, rs, re) in plan:
+        a, b = (rs, re) if reverse else (ss, se)
+        if b > a:
+            sbuf = feat[a:b].contiguous()
+            sends.append(sbuf)                       # keep alive until wait
+            reqs.append(dist.isend(sbuf, dst=q))
+    for r in reqs:
+        r.wait()
+    return recvs
+
+
+class _HaloSeq:
+    """RCCL p2p messages match by POSTING ORDER (no tags): every rank must
+    run its halo exchanges in the same order.  Forward order is program
+    order; backward order is enforced to be exactly reversed — a violation
+    raises instead of silently crossing buffers."""
+    fwd = 0
+    expect_bwd = None
+
+
+class HaloExchange(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, feat, plan):
+        ctx.plan = plan
+        ctx.seq = _HaloSeq.fwd
+        _HaloSeq.fwd += 1
+        recvs = _exchange(feat.detach(), plan, reverse=False)
+        out = feat.detach().clone()
+        for (q, ss, se, rs, re) in plan:
+            if re > rs:
+                out[rs:re] = recvs[q]
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        # the autograd engine drains its ready queue in descending
+        # sequence-number order, so live exchanges run in strictly
+        # DECREASING forward order on every rank (dead branches are skipped
+        # identically on all ranks — graph structure is rank-invariant).
+        if _HaloSeq.expect_bwd is None:
+            _HaloSeq.expect_bwd = _HaloSeq.fwd
+        if ctx.seq >= _HaloSeq.expect_bwd:
+            raise RuntimeError(
+                f"halo backward out of order: got seq {ctx.seq} after "
+                f"{_HaloSeq.expect_bwd} — rank-divergent exchange order")
+        _HaloSeq.expect_bwd = ctx.seq
+        plan = ctx.plan
+        g = grad.contiguous()
+        recvs = _exchange(g, plan, reverse=True)     # send ghost grads home
+        g = g.clone()
+        for (q, ss, se, rs, re) in plan:
+            if re > rs:
+                g[rs:re] = 0                         # ghosts aren't my vars
+        for (q, ss, se, rs, re) in plan:
+            if se > ss:
+                g[ss:se] += recvs[q]                 # peers' ghost grads
+        return g, None
+
+
+class SpmdEngine:
+    """Per-rank CHGNet E+F engine (the bench/production path)."""
+
+    def __init__(self, core: CHGNetCore, world: int, threads: int = 8,
+                 use_bond_graph: bool = True, device: Optional[str] = None,
+                 ops=None, graph_backend=None):
+        self.rank = dist.get_rank()
+        self.world = world
+        self.config = core.config
+        self.use_bond_graph = use_bond_graph and core.config.use_bond_graph
+        if device is None:
+            device = f"cuda:{torch.cuda.current_device()}"
+        self.device = torch.device(device)
+        self.core = deepcopy(core).to(self.device).eval()
+        self.ops = ops if ops is not None else default_ops_factory(self.device)
+        self.threads = threads
+        self.graph_backend = graph_backend
+        self.float_th = next(self.core.parameters()).dtype
+
+    # -- graph ------------------------------------------------------------
+
+    def build_graph(self, structure) -> Distributed:
+        return Distributed.create_distributed(
+            cart_coords=structure.cart_coords,
+            frac_coords=structure.frac_coords,
+            lattice_matrix=structure.lattice,
+            num_partitions=self.world, pbc=structure.pbc,
+            cutoff=self.config.cutoff,
+            three_body_cutoff=self.config.three_body_cutoff,
+            use_bond_graph=self.use_bond_graph,
+            num_threads=self.threads, backend=self.graph_backend)
+
+    # -- one E+F step ------------------------------------------------------
+
+    def step(self, structure, dist_info: Optional[Distributed] = None,
+             calc_stresses: bool = False):
+        """Returns (total_energy_scalar, owned_forces[n_owned,3]).
+
+        total_energy includes scale/shift + element refs (pes.py:109-113);
+        owned forces are fully assembled (reverse halo-add applied)."""
+        r, P = self.rank, self.world
+        cfg, core, ops, dev = self.config, self.core, self.ops, self.device
+        ft = self.float_th
+
+        if dist_info is None:
+            dist_info = self.build_graph(structure)
+        _HaloSeq.fwd = 0
+        _HaloSeq.expect_bwd = None
+        pd = PartitionData(dist_info, r, dev, self.use_bond_graph)
+        plan = halo_plan(dist_info.markers[r], r, P)
+        line_plan = halo_plan(dist_info.line_markers[r], r, P) \
+            if self.use_bond_graph else None
+        gids = np.asarray(dist_info.global_ids[r])
+        idx_dtype = torch.long if dev.type != "cuda" else torch.long
+
+        # ---- local geometry (per-rank; no GPU0 serialization point)
+        lat0 = torch.tensor(np.asarray(structure.lattice), dtype=ft, device=dev)
+        strain = lat0.new_zeros(3, 3)
+        if calc_stresses:
+            strain.requires_grad_(True)
+        lattice = lat0 @ (torch.eye(3, device=dev, dtype=ft) + strain)
+
+        frac_local = torch.tensor(np.asarray(structure.frac_coords)[gids],
+                                  dtype=ft, device=dev)
+        pos = frac_local @ lattice
+        if not pos.requires_grad:
+            pos.requires_grad_(True)
+        pos.retain_grad()
+
+        species_local = torch.tensor(np.asarray(structure.species)[gids],
+                                     dtype=torch.long, device=dev)
+
+        egids = np.asarray(dist_info.L2G_DE_mapping_list[r])
+        off_local = torch.tensor(np.asarray(dist_info.py_offsets)[egids],
+                                 dtype=ft, device=dev)
+        offshift = off_local @ lattice
+
+        bond_vec = ops.gather(pos, pd.dst, csr=(None, getattr(pd, "row_ptr", None))) \
+            + offshift - ops.gather(pos, pd.src,
+                                    csr=(getattr(pd, "src_perm", None),
+                                         getattr(pd, "src_row_ptr", None)))
+        bond_dist = torch.linalg.norm(bond_vec, dim=1)
+
+        bond_expansion = bond_expansion_from_dist(
+            bond_dist, core.rbf_freq_atom, cfg.cutoff, cfg.cutoff_exponent)
+
+        v = core.atom_embedding(species_local)
+        e = core.bond_embedding(bond_expansion)
+        w_ab = core.atom_bond_weights(bond_expansion)
+        w_bb = core.bond_bond_weights(bond_expansion)
+
+        use_bg = self.use_bond_graph
+        if use_bg:
+            # owned bond geometry from local edges; ghosts by one halo each
+            # (reference chgnet.py:129-164)
+            nd_dist = torch.zeros(pd.n_bonds, dtype=ft, device=dev).index_copy(
+                0, pd.map_ude, bond_dist[pd.map_de])
+            nd_vec = torch.zeros(pd.n_bonds, 3, dtype=ft, device=dev).index_copy(
+                0, pd.map_ude, bond_vec[pd.map_de])
+            nd_dist = HaloExchange.apply(nd_dist.unsqueeze(1), line_plan).squeeze(1)
+            nd_vec = HaloExchange.apply(nd_vec, line_plan)
+
+            exp3 = bond_expansion_from_dist(
+                nd_dist, core.rbf_freq_bond, cfg.three_body_cutoff,
+                cfg.cutoff_exponent)
+            theta = compute_theta(
+                ops.gather(nd_vec, pd.l_src, csr=pd.line_src_csr),
+                ops.gather(nd_vec, pd.l_dst, csr=pd.line_dst_csr))
+            a = core.angle_embedding(fourier_expansion(theta, core.angle_freq))
+            w_3b = core.threebody_bond_weights(exp3)
+            n = torch.zeros(pd.n_bonds, cfg.dim, dtype=ft, device=dev).index_copy(
+                0, pd.map_ude, e[pd.map_de])
+            n = HaloExchange.apply(n, line_plan)
+
+        d = cfg.dim
+
+        def atom_conv(layer_i, v, e):
+            blk = core.atom_convs[layer_i]
+            e = e + gated_mlp_split3(blk.edge_mlp, v, e, pd, ops, d) * w_bb
+            msg = gated_mlp_split3(blk.node_mlp, v, e, pd, ops, d) * w_ab
+            v = ops.scatter_edges(msg, pd, base=v)
+            return v, e
+
+        for layer_i in range(cfg.n_blocks - 1):           # chgnet.py:296-368
+            v, e = atom_conv(layer_i, v, e)
+            if use_bg:
+                n = n.index_copy(0, pd.map_ude, e[pd.map_de])   # edge_to_bond
+                n = HaloExchange.apply(n, line_plan)
+                v = HaloExchange.apply(v, plan)
+
+                blk = core.bond_convs[layer_i]
+                msg = gated_mlp_split4(blk.bond_mlp, n, a, v, pd, ops, d) * \
+                    ops.gather(w_3b, pd.l_src, csr=pd.line_src_csr)
+                n = ops.scatter_lines(msg, pd, base=n)
+                e = e.index_copy(0, pd.map_de, n[pd.map_ude])    # bond_to_edge
+
+                if layer_i < cfg.n_blocks - 2:
+                    # the LAST bond block's angle update is dead compute
+                    # (a never feeds the energy after it; the reference
+                    # still computes it, chgnet.py:353-368) — skip it and
+                    # the halo that feeds it
+                    n = HaloExchange.apply(n, line_plan)
+                    a = a + gated_mlp_split4(blk.angle_mlp, n, a, v, pd, ops, d)
+            else:
+                v = HaloExchange.apply(v, plan)
+
+        site_props = core.sitewise_readout(v)              # chgnet.py:391-398
+
+        v, e = atom_conv(-1, v, e)                         # final atom block
+        v = HaloExchange.apply(v, plan)
+
+        atom_e = core.final_layer(v)
+        n_owned = dist_info.num_owned_atoms(r)
+        e_local_raw = atom_e[:n_owned].sum()
+
+        refs_local = core.element_refs[species_local[:n_owned]].sum()
+
+        # forces: backward through std * E_local (mean/refs are pos-free)
+        loss = core.data_std * e_local_raw
+        grads = [pos, strain] if calc_stresses else [pos]
+        gv = torch.autograd.grad(loss, grads)
+        pos_grad = gv[0]
+
+        # reverse halo-add of position gradients -> exact owned forces
+        recvs = _exchange(pos_grad, plan, reverse=True)
+        pos_grad = pos_grad.clone()
+        for (q, ss, se, rs, re) in plan:
+            if se > ss:
+                pos_grad[ss:se] += recvs[q]
+        forces_owned = -pos_grad[:n_owned]
+
+        # scalar reductions for reporting
+        scal = torch.stack([e_local_raw.detach(), refs_local.detach()])
+        dist.all_reduce(scal)
+        total_e = core.data_std.detach() * scal[0] + core.data_mean.detach() \
+            + scal[1]
+
+        out = {"energy": total_e, "forces_owned": forces_owned,
+               "site_props_owned": site_props[:n_owned].detach(),
+               "n_owned": n_owned, "global_ids_owned": gids[:n_owned]}
+        if calc_stresses:
+            sg = gv[1].detach().clone()
+            dist.all_reduce(sg)
+            volume = float(np.abs(np.linalg.det(np.asarray(structure.lattice))))
+            out["stress"] = -sg / volume * -160.21766208
+        return out
