@@ -152,60 +152,40 @@ def main():
     if not use_bg:
         core.config.use_bond_graph = False
 
+    from distmlip_amd.runtime import SpmdEngine
     if world > 1:
-        from distmlip_amd.runtime import SpmdEngine
         torch.distributed.init_process_group("nccl")
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         torch.cuda.set_device(local_rank)
-        engine = SpmdEngine(core, world, threads=threads, use_bond_graph=use_bg)
+    engine = SpmdEngine(core.float(), world, threads=threads,
+                        use_bond_graph=use_bg)
 
-        def step():
-            return engine.step(s)
-    else:
-        model = CHGNet_Dist.from_existing(core, dtype=torch.float32)
-        model.enable_distributed_mode([0])
-        pot = Potential_Dist(model, calc_forces=True, num_threads=threads)
+    def step():
+        return engine.step(s)
 
-        def step():
-            E, F, _, _ = pot.forward(s)
-            return E, F
+    if world == 1 and args.breakdown:
+        stages = {}
 
-        if args.breakdown:
-            stages = {}
+        def mark(name, t0):
+            torch.cuda.synchronize()
+            stages.setdefault(name, []).append(time.time() - t0)
+            return time.time()
 
-            def step():  # noqa: F811
-                from distmlip_amd.dist import Distributed
+        def step():  # noqa: F811
+            t = time.time()
+            d = engine.build_graph(s)
+            t = mark("graph_cpu", t)
+            out = engine.step(s, dist_info=d)
+            t = mark("engine_step", t)
+            return out
 
-                def mark(name, t0):
-                    torch.cuda.synchronize()
-                    stages.setdefault(name, []).append(time.time() - t0)
-                    return time.time()
+        import atexit
 
-                t = time.time()
-                lattice_matrix = np.asarray(s.lattice, dtype=float)
-                d = Distributed.create_distributed(
-                    s.cart_coords, s.frac_coords, lattice_matrix, 1, s.pbc,
-                    6.0, 3.0, use_bond_graph=use_bg, num_threads=threads)
-                t = mark("graph_cpu", t)
-                model.set_local_species(d, s.species)
-                out = model.potential_forward_dist(
-                    d, s, lattice_matrix, False, True, False, None)
-                t = mark("h2d+forward", t)
-                node_types, positions, strain, (E, site) = out
-                E = model.core.data_std * E + model.core.data_mean
-                E = E + model.cores[0].element_refs[node_types].sum()
-                torch.autograd.backward(E)
-                F = -positions.grad
-                t = mark("backward", t)
-                return E, F
-
-            import atexit
-
-            def report():
-                for k, v in stages.items():
-                    print(f"# stage {k}: {1e3*sum(v)/len(v):.1f} ms avg over "
-                          f"{len(v)} calls", file=sys.stderr)
-            atexit.register(report)
+        def report():
+            for k, v in stages.items():
+                print(f"# stage {k}: {1e3*sum(v)/len(v):.1f} ms avg over "
+                      f"{len(v)} calls", file=sys.stderr)
+        atexit.register(report)
 
     timer = SegSumTimer().wrap()
 

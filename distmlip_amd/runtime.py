@@ -142,7 +142,9 @@ class SpmdEngine:
     def __init__(self, core: CHGNetCore, world: int, threads: int = 8,
                  use_bond_graph: bool = True, device: Optional[str] = None,
                  ops=None, graph_backend=None):
-        self.rank = dist.get_rank()
+        self.rank = dist.get_rank() if dist.is_initialized() else 0
+        assert world == 1 or dist.is_initialized(), \
+            "world > 1 needs an initialized torch.distributed process group"
         self.world = world
         self.config = core.config
         self.use_bond_graph = use_bond_graph and core.config.use_bond_graph
@@ -309,7 +311,8 @@ class SpmdEngine:
 
         # scalar reductions for reporting
         scal = torch.stack([e_local_raw.detach(), refs_local.detach()])
-        dist.all_reduce(scal)
+        if P > 1:
+            dist.all_reduce(scal)
         total_e = core.data_std.detach() * scal[0] + core.data_mean.detach() \
             + scal[1]
 
@@ -318,7 +321,8 @@ class SpmdEngine:
                "n_owned": n_owned, "global_ids_owned": gids[:n_owned]}
         if calc_stresses:
             sg = gv[1].detach().clone()
-            dist.all_reduce(sg)
+            if P > 1:
+                dist.all_reduce(sg)
             volume = float(np.abs(np.linalg.det(np.asarray(structure.lattice))))
             out["stress"] = -sg / volume * -160.21766208
         return out

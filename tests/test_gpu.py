@@ -117,6 +117,31 @@ def test_e2e_energy_forces_vs_oracle(P):
 
 
 @requires_gpu
+def test_spmd_engine_world1_vs_oracle():
+    """The bench/production path (SpmdEngine at world=1, per-rank geometry,
+    HIP kernels) against the fp64 oracle."""
+    from distmlip_amd.model import CHGNetCore
+    from distmlip_amd.runtime import SpmdEngine
+    from distmlip_amd.structures import diamond_si
+    from oracle.chgnet_ref import oracle_forward
+    from oracle.graph_ref import brute_force_neighbors
+
+    s = diamond_si((12, 2, 2), jitter=0.12, seed=2)
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    core = CHGNetCore.seeded(seed=0)
+    ref = oracle_forward(core.double(), s, g["src"], g["dst"], g["offsets"],
+                         g["within_bond_r"], dtype=torch.float64)
+    eng = SpmdEngine(core.float(), world=1, threads=4)
+    out = eng.step(s)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 5e-3 * max(
+        1.0, abs(ref["energy"].item()))
+    F = np.zeros((s.num_atoms, 3))
+    F[out["global_ids_owned"]] = out["forces_owned"].double().cpu().numpy()
+    dF = np.abs(F - ref["forces"].numpy()).max()
+    assert dF < 1e-4, f"engine force error {dF}"
+
+
+@requires_gpu
 def test_native_so_loaded():
     """Guard against silent eager fallback: the HIP extension must be the
     library the process actually loaded."""
