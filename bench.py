@@ -140,7 +140,9 @@ def main():
     n_gpus = max(args.gpus, world)
 
     use_bg = not args.no_bond_graph
-    threads = args.threads or max(1, (os.cpu_count() or 8) // max(world, 1))
+    # builder scaling saturates ~16-32 threads; oversubscribing a 256-thread
+    # host regresses 5x (NUMA + team spawn) — cap unless told otherwise
+    threads = args.threads or max(1, min(32, (os.cpu_count() or 8) // max(world, 1)))
 
     from distmlip_amd.chgnet import CHGNet_Dist
     from distmlip_amd.model import CHGNetCore

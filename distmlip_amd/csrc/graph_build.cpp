@@ -35,7 +35,20 @@
 
 #include <omp.h>
 
+#include <malloc.h>
+
 namespace {
+
+// The builder allocates multi-GB arrays per call; with glibc's default
+// mmap threshold every call mmaps fresh pages and every free munmaps them,
+// so each rebuild pays full page-fault cost (~1 s/GB).  Raise the
+// thresholds once so freed arena memory is REUSED across rebuilds.
+struct MallocTuning {
+    MallocTuning() {
+        mallopt(M_MMAP_THRESHOLD, 1 << 30);
+        mallopt(M_TRIM_THRESHOLD, 1 << 30);
+    }
+} g_malloc_tuning;
 
 thread_local std::string g_err;
 
