@@ -64,27 +64,27 @@ def halo_plan(markers, rank: int, P: int) -> List[Tuple[int, int, int, int, int]
 
 
 def _exchange(feat: torch.Tensor, plan, reverse: bool = False) -> dict:
-    """Post all irecvs then isends, wait; returns {peer: recv_buffer}.
-    reverse=True swaps roles (send my recv-slices, receive for my
-    send-slices) — the backward direction."""
-    recvs, reqs = {}, []
+    """Grouped p2p exchange (batch_isend_irecv: one RCCL group, deadlock-
+    free by construction); returns {peer: recv_buffer}.  reverse=True swaps
+    roles (send my recv-slices, receive for my send-slices) — the backward
+    direction."""
+    recvs, p2p, keep = {}, [], []
     for (q, ss, se, rs, re) in plan:
-        a, b = (rs, re) if reverse else (ss, se)
         c, d = (ss, se) if reverse else (rs, re)
         if d > c:
             buf = torch.empty((d - c,) + tuple(feat.shape[1:]),
                               dtype=feat.dtype, device=feat.device)
             recvs[q] = buf
-            reqs.append(dist.irecv(buf, src=q))
-    sends = []
+            p2p.append(dist.P2POp(dist.irecv, buf, q))
     for (q, ss, se, rs, re) in plan:
         a, b = (rs, re) if reverse else (ss, se)
         if b > a:
             sbuf = feat[a:b].contiguous()
-            sends.append(sbuf)                       # keep alive until wait
-            reqs.append(dist.isend(sbuf, dst=q))
-    for r in reqs:
-        r.wait()
+            keep.append(sbuf)                        # alive until wait
+            p2p.append(dist.P2POp(dist.isend, sbuf, q))
+    if p2p:
+        for r in dist.batch_isend_irecv(p2p):
+            r.wait()
     return recvs
 
 
