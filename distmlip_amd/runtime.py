@@ -141,7 +141,7 @@ class SpmdEngine:
 
     def __init__(self, core: CHGNetCore, world: int, threads: int = 8,
                  use_bond_graph: bool = True, device: Optional[str] = None,
-                 ops=None, graph_backend=None):
+                 ops=None, graph_backend=None, checkpoint: str = "auto"):
         self.rank = dist.get_rank() if dist.is_initialized() else 0
         assert world == 1 or dist.is_initialized(), \
             "world > 1 needs an initialized torch.distributed process group"
@@ -156,6 +156,12 @@ class SpmdEngine:
         self.threads = threads
         self.graph_backend = graph_backend
         self.float_th = next(self.core.parameters()).dtype
+        # activation checkpointing of the conv segments: at ~42M edges the
+        # saved [E,64] activations of 4 blocks would exceed the 288 GB HBM;
+        # recompute trades ~1/3 more kernel work for ~6x less live memory.
+        # Halo exchanges stay OUTSIDE checkpointed segments (a re-run of a
+        # comm op during backward would desynchronize the ranks).
+        self.checkpoint = checkpoint
 
     # -- graph ------------------------------------------------------------
 
@@ -253,13 +259,25 @@ class SpmdEngine:
             n = HaloExchange.apply(n, line_plan)
 
         d = cfg.dim
+        ckpt = self.checkpoint == "on" or (
+            self.checkpoint == "auto" and len(pd.src) > 10_000_000)
 
-        def atom_conv(layer_i, v, e):
-            blk = core.atom_convs[layer_i]
+        def _ck(fn, *args):
+            if ckpt:
+                return torch.utils.checkpoint.checkpoint(
+                    fn, *args, use_reentrant=False)
+            return fn(*args)
+
+        def atom_conv_body(layer_idx_t, v, e):
+            blk = core.atom_convs[int(layer_idx_t)]
             e = e + gated_mlp_split3(blk.edge_mlp, v, e, pd, ops, d) * w_bb
             msg = gated_mlp_split3(blk.node_mlp, v, e, pd, ops, d) * w_ab
             v = ops.scatter_edges(msg, pd, base=v)
             return v, e
+
+        def atom_conv(layer_i, v, e):
+            return _ck(atom_conv_body,
+                       torch.tensor(layer_i % cfg.n_blocks), v, e)
 
         for layer_i in range(cfg.n_blocks - 1):           # chgnet.py:296-368
             v, e = atom_conv(layer_i, v, e)
