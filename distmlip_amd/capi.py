@@ -66,21 +66,22 @@ class _PartView(ctypes.Structure):
                 ("n_owned_bonds", c_int64), ("n_lines", c_int64),
                 ("n_mapping", c_int64),
                 ("markers", POINTER(c_int64)), ("global_ids", POINTER(c_int64)),
-                ("src_local", POINTER(c_int64)), ("dst_local", POINTER(c_int64)),
+                ("src_local", POINTER(c_int32)), ("dst_local", POINTER(c_int32)),
                 ("edge_gids", POINTER(c_int64)),
                 ("line_markers", POINTER(c_int64)),
-                ("line_src", POINTER(c_int64)), ("line_dst", POINTER(c_int64)),
-                ("line_center", POINTER(c_int64)),
+                ("line_src", POINTER(c_int32)), ("line_dst", POINTER(c_int32)),
+                ("line_center", POINTER(c_int32)),
                 ("map_de", POINTER(c_int64)), ("map_ude", POINTER(c_int64)),
                 ("bde_edge_gids", POINTER(c_int64)),
-                ("row_ptr", POINTER(c_int64)),
-                ("src_perm", POINTER(c_int64)),
-                ("src_row_ptr", POINTER(c_int64)),
-                ("line_row_ptr", POINTER(c_int64)),
-                ("line_src_perm", POINTER(c_int64)),
-                ("line_src_row_ptr", POINTER(c_int64)),
-                ("center_perm", POINTER(c_int64)),
-                ("center_row_ptr", POINTER(c_int64))]
+                ("row_ptr", POINTER(c_int32)),
+                ("src_perm", POINTER(c_int32)),
+                ("src_row_ptr", POINTER(c_int32)),
+                ("line_row_ptr", POINTER(c_int32)),
+                ("line_src_perm", POINTER(c_int32)),
+                ("line_src_row_ptr", POINTER(c_int32)),
+                ("center_perm", POINTER(c_int32)),
+                ("center_row_ptr", POINTER(c_int32)),
+                ("offsets_i8", POINTER(ctypes.c_int8))]
 
 
 class _Owner:
@@ -97,10 +98,14 @@ class _Owner:
             pass
 
 
+_CTYPE_OF = {np.int64: c_int64, np.float64: c_double,
+             np.int32: c_int32, np.int8: ctypes.c_int8}
+
+
 def _as_np(ptr, count, nptype, owner, shape=None):
     if count == 0 or not ptr:
         return np.zeros(shape if shape else (0,), dtype=nptype)
-    ctype = c_int64 if nptype == np.int64 else c_double
+    ctype = _CTYPE_OF[nptype]
     buf = (ctype * count).from_address(
         ctypes.cast(ptr, c_void_p).value)
     buf._owner = owner  # keepalive chain (ctypes instances allow attributes)
@@ -149,29 +154,31 @@ def build_graph(frac_coords, lattice, pbc, cutoff, bond_cutoff, tol,
         parts.append({
             "markers": _as_np(pv.markers, 2 * num_partitions + 1, np.int64, owner),
             "global_ids": _as_np(pv.global_ids, pv.n_nodes, np.int64, owner),
-            "src_local": _as_np(pv.src_local, pv.n_edges, np.int64, owner),
-            "dst_local": _as_np(pv.dst_local, pv.n_edges, np.int64, owner),
+            "src_local": _as_np(pv.src_local, pv.n_edges, np.int32, owner),
+            "dst_local": _as_np(pv.dst_local, pv.n_edges, np.int32, owner),
             "edge_gids": _as_np(pv.edge_gids, pv.n_edges, np.int64, owner),
             "line_markers": _as_np(pv.line_markers, 2 * num_partitions + 1,
                                    np.int64, owner) if use_bond_graph else None,
-            "line_src": _as_np(pv.line_src, pv.n_lines, np.int64, owner),
-            "line_dst": _as_np(pv.line_dst, pv.n_lines, np.int64, owner),
-            "line_center": _as_np(pv.line_center, pv.n_lines, np.int64, owner),
+            "line_src": _as_np(pv.line_src, pv.n_lines, np.int32, owner),
+            "line_dst": _as_np(pv.line_dst, pv.n_lines, np.int32, owner),
+            "line_center": _as_np(pv.line_center, pv.n_lines, np.int32, owner),
             "map_de": _as_np(pv.map_de, pv.n_mapping, np.int64, owner),
             "map_ude": _as_np(pv.map_ude, pv.n_mapping, np.int64, owner),
             "bde_edge_gids": _as_np(pv.bde_edge_gids, pv.n_bonds, np.int64, owner),
             "n_bonds": pv.n_bonds,
             "n_owned_bonds": pv.n_owned_bonds,
-            "row_ptr": _as_np(pv.row_ptr, pv.n_nodes + 1, np.int64, owner),
-            "src_perm": _as_np(pv.src_perm, pv.n_edges, np.int64, owner),
-            "src_row_ptr": _as_np(pv.src_row_ptr, pv.n_nodes + 1, np.int64, owner),
-            "line_row_ptr": _as_np(pv.line_row_ptr, pv.n_bonds + 1, np.int64, owner),
-            "line_src_perm": _as_np(pv.line_src_perm, pv.n_lines, np.int64, owner),
+            "row_ptr": _as_np(pv.row_ptr, pv.n_nodes + 1, np.int32, owner),
+            "src_perm": _as_np(pv.src_perm, pv.n_edges, np.int32, owner),
+            "src_row_ptr": _as_np(pv.src_row_ptr, pv.n_nodes + 1, np.int32, owner),
+            "line_row_ptr": _as_np(pv.line_row_ptr, pv.n_bonds + 1, np.int32, owner),
+            "line_src_perm": _as_np(pv.line_src_perm, pv.n_lines, np.int32, owner),
             "line_src_row_ptr": _as_np(pv.line_src_row_ptr, pv.n_bonds + 1,
-                                       np.int64, owner),
-            "center_perm": _as_np(pv.center_perm, pv.n_lines, np.int64, owner),
+                                       np.int32, owner),
+            "center_perm": _as_np(pv.center_perm, pv.n_lines, np.int32, owner),
             "center_row_ptr": _as_np(pv.center_row_ptr, pv.n_nodes + 1,
-                                     np.int64, owner),
+                                     np.int32, owner),
+            "offsets_i8": _as_np(pv.offsets_i8, 3 * pv.n_edges, np.int8, owner,
+                                 (pv.n_edges, 3)),
         })
     return owner, g, parts
 

@@ -55,28 +55,39 @@ def _split_first(lin: nn.Linear, parts: List[int]):
     return ws, lin.bias
 
 
-def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int):
-    """GatedMLP over cat(v[src], v[dst], e) via split-linear + gather_add3."""
-    (wc_s, wc_d, wc_e), bc = _split_first(mlp.core1, [d, d, d])
-    (wg_s, wg_d, wg_e), bg = _split_first(mlp.gate1, [d, d, d])
-    zc = ops.gather_add3(v @ wc_s.t(), v @ wc_d.t(), e @ wc_e.t() + bc, pd)
-    zg = ops.gather_add3(v @ wg_s.t(), v @ wg_d.t(), e @ wg_e.t() + bg, pd)
-    core = F.silu(mlp.core2(F.silu(zc)))
-    gate = torch.sigmoid(mlp.gate2(F.silu(zg)))
-    return core * gate
+def _cg_weight(mlp: GatedMLP):
+    """[core1 ; gate1] stacked weight [2h, in] + bias [2h] (one GEMM feeds
+    both branches; the input is read once)."""
+    w = torch.cat([mlp.core1.weight, mlp.gate1.weight], dim=0)
+    b = torch.cat([mlp.core1.bias, mlp.gate1.bias], dim=0)
+    return w, b
 
 
-def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int):
-    """GatedMLP over cat(n[l_src], n[l_dst], a, v[center])."""
-    (wc_1, wc_2, wc_a, wc_v), bc = _split_first(mlp.core1, [d, d, d, d])
-    (wg_1, wg_2, wg_a, wg_v), bg = _split_first(mlp.gate1, [d, d, d, d])
-    zc = ops.gather_add4(n @ wc_1.t(), n @ wc_2.t(), a @ wc_a.t() + bc,
-                         v @ wc_v.t(), pd)
-    zg = ops.gather_add4(n @ wg_1.t(), n @ wg_2.t(), a @ wg_a.t() + bg,
-                         v @ wg_v.t(), pd)
-    core = F.silu(mlp.core2(F.silu(zc)))
-    gate = torch.sigmoid(mlp.gate2(F.silu(zg)))
-    return core * gate
+def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
+    """GatedMLP over cat(v[src], v[dst], e) via split-linear + one 2h-wide
+    gather_add3, finished by the fused gated-combine epilogue:
+    returns base + silu(core2(silu(z_c))) * sigmoid(gate2(silu(z_g))) * w."""
+    wcg, bcg = _cg_weight(mlp)
+    ws, wd, we = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:]
+    z = ops.gather_add3(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
+    h = F.silu(z)
+    c = mlp.core2(h[:, :d])
+    g = mlp.gate2(h[:, d:])
+    return ops.gated_combine(c, g, w, base)
+
+
+def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
+                     base=None):
+    """GatedMLP over cat(n[l_src], n[l_dst], a, v[center]), same structure."""
+    wcg, bcg = _cg_weight(mlp)
+    w1, w2, wa, wv = (wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:3 * d],
+                      wcg[:, 3 * d:])
+    z = ops.gather_add4(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
+                        v @ wv.t(), pd)
+    h = F.silu(z)
+    c = mlp.core2(h[:, :d])
+    g = mlp.gate2(h[:, d:])
+    return ops.gated_combine(c, g, w, base)
 
 
 class PartitionData:
@@ -303,10 +314,11 @@ class CHGNet_Dist(nn.Module):
         def atom_conv(layer_i):
             for p in range(P):
                 blk, pd, ops = self.cores[p].atom_convs[layer_i], parts[p], self.ops[p]
-                e_list[p] = e_list[p] + gated_mlp_split3(
-                    blk.edge_mlp, v_list[p], e_list[p], pd, ops, d) * w_bb[p]
+                e_list[p] = gated_mlp_split3(
+                    blk.edge_mlp, v_list[p], e_list[p], pd, ops, d,
+                    w=w_bb[p], base=e_list[p])
                 msg = gated_mlp_split3(
-                    blk.node_mlp, v_list[p], e_list[p], pd, ops, d) * w_ab[p]
+                    blk.node_mlp, v_list[p], e_list[p], pd, ops, d, w=w_ab[p])
                 v_list[p] = ops.scatter_edges(msg, pd, base=v_list[p])
 
         for layer_i in range(cfg.n_blocks - 1):          # chgnet.py:296-368
@@ -322,8 +334,8 @@ class CHGNet_Dist(nn.Module):
                     blk, pd, ops = self.cores[p].bond_convs[layer_i], parts[p], self.ops[p]
                     msg = gated_mlp_split4(
                         blk.bond_mlp, n_list[p], a_list[p], v_list[p],
-                        pd, ops, d) * \
-                        ops.gather(w_3b[p], pd.l_src, csr=pd.line_src_csr)
+                        pd, ops, d,
+                        w=ops.gather(w_3b[p], pd.l_src, csr=pd.line_src_csr))
                     n_list[p] = ops.scatter_lines(msg, pd, base=n_list[p])
                     dist_info.bond_to_edge(n_list, e_list, p)
 
@@ -334,9 +346,9 @@ class CHGNet_Dist(nn.Module):
                     dist_info.bond_transfer(n_list)
                     for p in range(P):                   # angle pass chgnet.py:353-368
                         blk, pd, ops = self.cores[p].bond_convs[layer_i], parts[p], self.ops[p]
-                        a_list[p] = a_list[p] + gated_mlp_split4(
+                        a_list[p] = gated_mlp_split4(
                             blk.angle_mlp, n_list[p], a_list[p], v_list[p],
-                            pd, ops, d)
+                            pd, ops, d, base=a_list[p])
             else:
                 dist_info.atom_transfer(v_list)
 

@@ -70,24 +70,31 @@ template <class T> using rvec = std::vector<T, NoInit<T>>;
 
 struct Part {
     std::vector<int64_t> markers, line_markers, map_de, map_ude;
-    rvec<int64_t> global_ids, src_local, dst_local, edge_gids;
-    rvec<int64_t> line_src, line_dst, line_center, bde_edge_gids;
-    rvec<int64_t> row_ptr, src_perm, src_row_ptr;
-    rvec<int64_t> line_row_ptr, line_src_perm, line_src_row_ptr;
-    rvec<int64_t> center_perm, center_row_ptr;
+    rvec<int64_t> global_ids, edge_gids, bde_edge_gids;
+    // hot device-bound arrays are int32 (n < 2^31) so the Python layer
+    // uploads them without a host-side conversion pass
+    rvec<int32_t> src_local, dst_local, line_src, line_dst, line_center;
+    rvec<int32_t> row_ptr, src_perm, src_row_ptr;
+    rvec<int32_t> line_row_ptr, line_src_perm, line_src_row_ptr;
+    rvec<int32_t> center_perm, center_row_ptr;
+    rvec<int8_t> off_i8;     // [n_edges*3] integer images of dst (|img|<=127)
     int64_t n_owned = 0, n_owned_bonds = 0;
 };
 
 // stable counting sort of [0..m) by key[i] in [0..nkeys); emits perm and
 // CSR row_ptr[nkeys+1]
-void counting_csr(const int64_t* key, int64_t m, int64_t nkeys,
-                  rvec<int64_t>& perm, rvec<int64_t>& rptr) {
-    rptr.assign(nkeys + 1, 0);
-    for (int64_t i = 0; i < m; ++i) ++rptr[key[i] + 1];
-    for (int64_t k = 0; k < nkeys; ++k) rptr[k + 1] += rptr[k];
+template <class K>
+void counting_csr(const K* key, int64_t m, int64_t nkeys,
+                  rvec<int32_t>& perm, rvec<int32_t>& rptr) {
+    std::vector<int64_t> cnt(nkeys + 1, 0);
+    for (int64_t i = 0; i < m; ++i) ++cnt[key[i] + 1];
+    for (int64_t k = 0; k < nkeys; ++k) cnt[k + 1] += cnt[k];
+    rptr.resize(nkeys + 1);
+    for (int64_t k = 0; k <= nkeys; ++k) rptr[k] = (int32_t)cnt[k];
     perm.resize(m);
     std::vector<int64_t> fill(nkeys, 0);
-    for (int64_t i = 0; i < m; ++i) perm[rptr[key[i]] + fill[key[i]]++] = i;
+    for (int64_t i = 0; i < m; ++i)
+        perm[cnt[key[i]] + fill[key[i]]++] = (int32_t)i;
 }
 
 }  // namespace
@@ -446,15 +453,19 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
                         ++o;
                     }
             }
-            rvec<int64_t> eperm;
+            rvec<int32_t> eperm;
             counting_csr(tdst.data(), Ep, Nn, eperm, pt.row_ptr);
             pt.src_local.resize(Ep); pt.dst_local.resize(Ep); pt.edge_gids.resize(Ep);
+            pt.off_i8.resize(3 * Ep);
 #pragma omp parallel for num_threads(T) schedule(static)
             for (int64_t i = 0; i < Ep; ++i) {
                 const int64_t j = eperm[i];
-                pt.src_local[i] = tsrc[j];
-                pt.dst_local[i] = tdst[j];
+                pt.src_local[i] = (int32_t)tsrc[j];
+                pt.dst_local[i] = (int32_t)tdst[j];
                 pt.edge_gids[i] = tgid[j];
+                pt.off_i8[3 * i] = (int8_t)g->offsets[3 * tgid[j]];
+                pt.off_i8[3 * i + 1] = (int8_t)g->offsets[3 * tgid[j] + 1];
+                pt.off_i8[3 * i + 2] = (int8_t)g->offsets[3 * tgid[j] + 2];
                 if (use_bond_graph) g2l_edge[tgid[j]] = i;
             }
             counting_csr(pt.src_local.data(), Ep, Nn, pt.src_perm, pt.src_row_ptr);
@@ -547,15 +558,15 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
             }
         }
         // l_dst-sorted line arrays + CSRs (scatter layouts for the kernels)
-        rvec<int64_t> lperm;
+        rvec<int32_t> lperm;
         counting_csr(tld.data(), L, B, lperm, pt.line_row_ptr);
         pt.line_src.resize(L); pt.line_dst.resize(L); pt.line_center.resize(L);
 #pragma omp parallel for num_threads(nthreads) schedule(static)
         for (int64_t i = 0; i < L; ++i) {
             const int64_t j = lperm[i];
-            pt.line_src[i] = tls[j];
-            pt.line_dst[i] = tld[j];
-            pt.line_center[i] = tlc[j];
+            pt.line_src[i] = (int32_t)tls[j];
+            pt.line_dst[i] = (int32_t)tld[j];
+            pt.line_center[i] = (int32_t)tlc[j];
         }
         counting_csr(pt.line_src.data(), L, B, pt.line_src_perm,
                      pt.line_src_row_ptr);
@@ -636,6 +647,7 @@ int dm_graph_partition_view(const dm_graph* g, int32_t partition,
     out->line_src_row_ptr = pt.line_src_row_ptr.empty() ? nullptr : pt.line_src_row_ptr.data();
     out->center_perm = pt.center_perm.data();
     out->center_row_ptr = pt.center_row_ptr.empty() ? nullptr : pt.center_row_ptr.data();
+    out->offsets_i8 = pt.off_i8.data();
     return 0;
 }
 

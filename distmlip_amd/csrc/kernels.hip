@@ -213,6 +213,47 @@ __global__ void k_seg_sum_gather_s(const float* __restrict__ msg,
     }
 }
 
+__device__ __forceinline__ float sigf(float x) {
+    return 1.0f / (1.0f + __expf(-x));
+}
+
+// out = base + silu(c) * sigmoid(g) * w   (gated-MLP epilogue, fused)
+__global__ void k_gated_combine_fwd(const float* __restrict__ c,
+                                    const float* __restrict__ g,
+                                    const float* __restrict__ w,
+                                    const float* __restrict__ base,
+                                    float* __restrict__ out, int64_t total) {
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const float cv = c[t];
+        float r = cv * sigf(cv) * sigf(g[t]);
+        if (w) r *= w[t];
+        if (base) r += base[t];
+        out[t] = r;
+    }
+}
+
+__global__ void k_gated_combine_bwd(const float* __restrict__ go,
+                                    const float* __restrict__ c,
+                                    const float* __restrict__ g,
+                                    const float* __restrict__ w,
+                                    float* __restrict__ dc,
+                                    float* __restrict__ dg,
+                                    float* __restrict__ dw, int64_t total) {
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const float gov = go[t];
+        const float cv = c[t], gv = g[t];
+        const float sc = sigf(cv), sg = sigf(gv);
+        const float silu_c = cv * sc;
+        const float wv = w ? w[t] : 1.0f;
+        // d silu(c)/dc = sigmoid(c) * (1 + c * (1 - sigmoid(c)))
+        dc[t] = gov * wv * sg * (sc * (1.0f + cv * (1.0f - sc)));
+        dg[t] = gov * wv * silu_c * (sg * (1.0f - sg));
+        if (dw) dw[t] = gov * silu_c * sg;
+    }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -292,6 +333,26 @@ int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
         k_seg_sum_gather_s<<<nblocks(N * D, BLOCK), BLOCK, 0, s>>>(
             msg, perm, row_ptr, base, out, N, (int32_t)D);
     }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_gated_combine_fwd_f32(const float* c, const float* g, const float* w,
+                             const float* base, float* out, int64_t total,
+                             uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    k_gated_combine_fwd<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
+        c, g, w, base, out, total);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_gated_combine_bwd_f32(const float* go, const float* c, const float* g,
+                             const float* w, float* dc, float* dg, float* dw,
+                             int64_t total, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    k_gated_combine_bwd<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
+        go, c, g, w, dc, dg, dw, total);
     DM_CHECK_LAUNCH();
     return 0;
 }

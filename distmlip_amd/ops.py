@@ -42,6 +42,12 @@ def hip_lib() -> ctypes.CDLL:
         lib.dm_seg_sum_gather_f32.restype = c_int32
         lib.dm_seg_sum_gather_f32.argtypes = [fp, ip, ip, fp, fp, c_int64,
                                               c_int64, c_uint64]
+        lib.dm_gated_combine_fwd_f32.restype = c_int32
+        lib.dm_gated_combine_fwd_f32.argtypes = [fp, fp, fp, fp, fp, c_int64,
+                                                 c_uint64]
+        lib.dm_gated_combine_bwd_f32.restype = c_int32
+        lib.dm_gated_combine_bwd_f32.argtypes = [fp, fp, fp, fp, fp, fp, fp,
+                                                 c_int64, c_uint64]
         lib.dm_hip_last_error.restype = c_char_p
         _hip_lib = lib
     return _hip_lib
@@ -188,6 +194,41 @@ class _SegSum(torch.autograd.Function):
         return gmsg, None, None, None, gbase
 
 
+class _GatedCombine(torch.autograd.Function):
+    """out = base + silu(c) * sigmoid(g) * w — fused gated-MLP epilogue."""
+
+    @staticmethod
+    def forward(ctx, c, g, w, base):
+        _chk_f32(c, g, w, base)
+        ctx.save_for_backward(c, g) if w is None else \
+            ctx.save_for_backward(c, g, w)
+        ctx.has_w = w is not None
+        ctx.has_base = base is not None
+        out = torch.empty_like(c)
+        _check(hip_lib().dm_gated_combine_fwd_f32(
+            _fp(c), _fp(g), _fp(w) if w is not None else None,
+            _fp(base) if base is not None else None, _fp(out), c.numel(),
+            _stream()), "dm_gated_combine_fwd_f32")
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        if ctx.has_w:
+            c, g, w = ctx.saved_tensors
+        else:
+            c, g = ctx.saved_tensors
+            w = None
+        go = go.contiguous()
+        dc = torch.empty_like(c)
+        dg = torch.empty_like(g)
+        dw = torch.empty_like(w) if ctx.has_w else None
+        _check(hip_lib().dm_gated_combine_bwd_f32(
+            _fp(go), _fp(c), _fp(g), _fp(w) if w is not None else None,
+            _fp(dc), _fp(dg), _fp(dw) if dw is not None else None,
+            c.numel(), _stream()), "dm_gated_combine_bwd_f32")
+        return dc, dg, dw, (go if ctx.has_base else None)
+
+
 class HipOps:
     """Product ops backend (see ops_base.OpsBackend)."""
 
@@ -213,3 +254,9 @@ class HipOps:
 
     def scatter_lines(self, msg, pd, base=None):
         return _SegSum.apply(msg, pd.l_dst, pd.line_row_ptr, pd.n_bonds, base)
+
+    def gated_combine(self, c, g, w=None, base=None):
+        return _GatedCombine.apply(
+            c.contiguous(), g.contiguous(),
+            w.contiguous() if w is not None else None,
+            base.contiguous() if base is not None else None)

@@ -79,13 +79,15 @@ typedef struct {
     int64_t n_mapping;          /* bond_mapping pairs (owned BDEs) */
     const int64_t* markers;        /* [2P+1] region starts (python appends total) */
     const int64_t* global_ids;     /* [n_nodes] local -> global atom id */
-    const int64_t* src_local;      /* [n_edges] */
-    const int64_t* dst_local;      /* [n_edges] */
+    const int32_t* src_local;      /* [n_edges] (int32: device-bound arrays
+                                      are 32-bit so the host uploads them
+                                      without a conversion pass) */
+    const int32_t* dst_local;      /* [n_edges] */
     const int64_t* edge_gids;      /* [n_edges] local -> global edge id (L2G) */
     const int64_t* line_markers;   /* [2P+1] */
-    const int64_t* line_src;       /* [n_lines] local BDE ids */
-    const int64_t* line_dst;       /* [n_lines] */
-    const int64_t* line_center;    /* [n_lines] LOCAL atom id of center */
+    const int32_t* line_src;       /* [n_lines] local BDE ids */
+    const int32_t* line_dst;       /* [n_lines] */
+    const int32_t* line_center;    /* [n_lines] LOCAL atom id of center */
     const int64_t* map_de;         /* [n_mapping] local edge id */
     const int64_t* map_ude;        /* [n_mapping] local BDE id */
     const int64_t* bde_edge_gids;  /* [n_bonds] global edge id per BDE,
@@ -98,14 +100,16 @@ typedef struct {
      * scatter-add is a contiguous segmented reduction; every other
      * scatter direction gets a permutation CSR (deterministic backward,
      * no atomics).  Line edges are likewise emitted l_dst-sorted. */
-    const int64_t* row_ptr;            /* [n_nodes+1] CSR over dst_local   */
-    const int64_t* src_perm;           /* [n_edges]  edge ids sorted by src */
-    const int64_t* src_row_ptr;        /* [n_nodes+1]                      */
-    const int64_t* line_row_ptr;       /* [n_bonds+1] CSR over line_dst    */
-    const int64_t* line_src_perm;      /* [n_lines] line ids sorted by line_src */
-    const int64_t* line_src_row_ptr;   /* [n_bonds+1]                      */
-    const int64_t* center_perm;        /* [n_lines] line ids sorted by center atom */
-    const int64_t* center_row_ptr;     /* [n_nodes+1]                      */
+    const int32_t* row_ptr;            /* [n_nodes+1] CSR over dst_local   */
+    const int32_t* src_perm;           /* [n_edges]  edge ids sorted by src */
+    const int32_t* src_row_ptr;        /* [n_nodes+1]                      */
+    const int32_t* line_row_ptr;       /* [n_bonds+1] CSR over line_dst    */
+    const int32_t* line_src_perm;      /* [n_lines] line ids sorted by line_src */
+    const int32_t* line_src_row_ptr;   /* [n_bonds+1]                      */
+    const int32_t* center_perm;        /* [n_lines] line ids sorted by center atom */
+    const int32_t* center_row_ptr;     /* [n_nodes+1]                      */
+    const int8_t*  offsets_i8;         /* [n_edges,3] integer image of dst
+                                          per LOCAL edge (|image| <= 127)  */
 } dm_partition_view;
 
 int dm_graph_partition_view(const dm_graph* g, int32_t partition,

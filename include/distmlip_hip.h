@@ -60,6 +60,16 @@ int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
                           const int32_t* row_ptr, const float* base,
                           float* out, int64_t N, int64_t D, uint64_t stream);
 
+/* out = (base?base:0) + silu(c) * sigmoid(g) * (w?w:1) — the gated-MLP
+ * epilogue (core activation x gate x shared message weight x residual),
+ * fused from ~5 eager passes.  bwd emits dc, dg and (if w) dw. */
+int dm_gated_combine_fwd_f32(const float* c, const float* g, const float* w,
+                             const float* base, float* out, int64_t total,
+                             uint64_t stream);
+int dm_gated_combine_bwd_f32(const float* go, const float* c, const float* g,
+                             const float* w, float* dc, float* dg, float* dw,
+                             int64_t total, uint64_t stream);
+
 const char* dm_hip_last_error(void);
 
 #ifdef __cplusplus

@@ -209,3 +209,9 @@ class CpuRefOps:
 
     def scatter_lines(self, msg, pd, base=None):
         return self._scatter(msg, pd.l_dst, pd.n_bonds, base)
+
+    def gated_combine(self, c, g, w=None, base=None):
+        out = torch.nn.functional.silu(c) * torch.sigmoid(g)
+        if w is not None:
+            out = out * w
+        return out if base is None else base + out

@@ -86,6 +86,41 @@ class TestKernels:
         assert torch.allclose(ze.grad, ze2.grad)
 
 
+@requires_gpu
+def test_gated_combine_fwd_bwd():
+    from distmlip_amd.ops import _GatedCombine
+    torch.manual_seed(1)
+    dev = torch.device("cuda:0")
+    E, D = 4000, 64
+    for has_w, has_base in [(True, True), (True, False), (False, True),
+                            (False, False)]:
+        c = torch.randn(E, D, device=dev, requires_grad=True)
+        g = torch.randn(E, D, device=dev, requires_grad=True)
+        w = torch.randn(E, D, device=dev, requires_grad=True) if has_w else None
+        base = torch.randn(E, D, device=dev, requires_grad=True) if has_base else None
+        out = _GatedCombine.apply(c, g, w, base)
+        ref = torch.nn.functional.silu(c.detach().clone().requires_grad_(True))
+        c2 = c.detach().clone().requires_grad_(True)
+        g2 = g.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True) if has_w else None
+        b2 = base.detach().clone().requires_grad_(True) if has_base else None
+        r = torch.nn.functional.silu(c2) * torch.sigmoid(g2)
+        if has_w:
+            r = r * w2
+        if has_base:
+            r = b2 + r
+        assert torch.allclose(out, r, atol=2e-6), (has_w, has_base)
+        go = torch.randn_like(out)
+        out.backward(go)
+        r.backward(go)
+        assert torch.allclose(c.grad, c2.grad, atol=2e-5)
+        assert torch.allclose(g.grad, g2.grad, atol=2e-5)
+        if has_w:
+            assert torch.allclose(w.grad, w2.grad, atol=2e-5)
+        if has_base:
+            assert torch.allclose(base.grad, b2.grad)
+
+
 def _gpu_model(core, P, devices=None):
     from distmlip_amd.chgnet import CHGNet_Dist
     model = CHGNet_Dist.from_existing(core, dtype=torch.float32)
