@@ -28,6 +28,9 @@ import numpy as np
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+# avoid allocator fragmentation at the 1M-atom workload (10+ GiB blocks)
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import torch  # noqa: E402
 
 HBM_PEAK_BYTES = 8.0e12  # MI355X spec peak (MI355X_MICROARCH.md); measured

@@ -213,8 +213,146 @@ __global__ void k_seg_sum_gather_s(const float* __restrict__ msg,
     }
 }
 
+// rbf_k(d) and d(rbf_env)/dd helpers for the fused geometry kernels.
+// env is the reference's polynomial cutoff applied to the RBF VALUE
+// (chgnet.py:119-121); both value and derivative are exact fp32 forms of
+// distmlip_amd.model.bond_expansion_from_dist.
+template <int MAXR>
+__device__ __forceinline__ void rbf_env_row(float bd, const float* freqs,
+                                            float cutoff, int pexp, int nrbf,
+                                            float* out) {
+    const float norm = sqrtf(2.0f / cutoff);
+    const float inv_d = 1.0f / bd;
+    const float e = (float)pexp;
+    const float c1 = -(e + 1.0f) * (e + 2.0f) * 0.5f;
+    const float c2 = e * (e + 2.0f);
+    const float c3 = -e * (e + 1.0f) * 0.5f;
+    for (int k = 0; k < nrbf; ++k) {
+        const float rbf = norm * sinf(freqs[k] * bd / cutoff) * inv_d;
+        const float r = rbf / cutoff;
+        const float rp = powf(r, e);
+        const float env = (rbf <= cutoff)
+            ? 1.0f + c1 * rp + c2 * rp * r + c3 * rp * r * r : 0.0f;
+        out[k] = env * rbf;
+    }
+}
+
+template <int MAXR>
+__device__ __forceinline__ float rbf_env_grad_dot(float bd, const float* freqs,
+                                                  float cutoff, int pexp,
+                                                  int nrbf,
+                                                  const float* go_row) {
+    // sum_k go[k] * d(env(rbf_k)*rbf_k)/d(bd)
+    const float norm = sqrtf(2.0f / cutoff);
+    const float inv_d = 1.0f / bd;
+    const float e = (float)pexp;
+    const float c1 = -(e + 1.0f) * (e + 2.0f) * 0.5f;
+    const float c2 = e * (e + 2.0f);
+    const float c3 = -e * (e + 1.0f) * 0.5f;
+    float acc = 0.0f;
+    for (int k = 0; k < nrbf; ++k) {
+        const float a = freqs[k] / cutoff;
+        float sn, cs;
+        sincosf(a * bd, &sn, &cs);
+        const float rbf = norm * sn * inv_d;
+        const float drbf = norm * (cs * a * inv_d - sn * inv_d * inv_d);
+        const float r = rbf / cutoff;
+        const float rpm1 = powf(r, e - 1.0f);
+        const float rp = rpm1 * r;
+        float env, denv;
+        if (rbf <= cutoff) {
+            env = 1.0f + c1 * rp + c2 * rp * r + c3 * rp * r * r;
+            denv = (e * c1 * rpm1 + (e + 1.0f) * c2 * rp
+                    + (e + 2.0f) * c3 * rp * r) / cutoff;
+        } else { env = 0.0f; denv = 0.0f; }
+        acc += go_row[k] * (env + denv * rbf) * drbf;
+    }
+    return acc;
+}
+
+__global__ void k_edge_geom_rbf_fwd(const float* __restrict__ pos,
+                                    const int32_t* __restrict__ src,
+                                    const int32_t* __restrict__ dst,
+                                    const float* __restrict__ offshift,
+                                    const float* __restrict__ freqs,
+                                    float cutoff, int pexp, int nrbf,
+                                    float* __restrict__ bv,
+                                    float* __restrict__ bd,
+                                    float* __restrict__ exp_out, int64_t E) {
+    __shared__ float fsh[16];
+    if (threadIdx.x < nrbf) fsh[threadIdx.x] = freqs[threadIdx.x];
+    __syncthreads();
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < E; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t s3 = (int64_t)src[t] * 3, d3 = (int64_t)dst[t] * 3;
+        const float x = pos[d3] + offshift[3 * t] - pos[s3];
+        const float y = pos[d3 + 1] + offshift[3 * t + 1] - pos[s3 + 1];
+        const float z = pos[d3 + 2] + offshift[3 * t + 2] - pos[s3 + 2];
+        bv[3 * t] = x; bv[3 * t + 1] = y; bv[3 * t + 2] = z;
+        const float d = sqrtf(x * x + y * y + z * z);
+        bd[t] = d;
+        float row[16];
+        rbf_env_row<16>(d, fsh, cutoff, pexp, nrbf, row);
+        for (int k = 0; k < nrbf; ++k) exp_out[t * nrbf + k] = row[k];
+    }
+}
+
+__global__ void k_edge_geom_rbf_bwd(const float* __restrict__ go_bv,
+                                    const float* __restrict__ go_bd,
+                                    const float* __restrict__ go_exp,
+                                    const float* __restrict__ bv,
+                                    const float* __restrict__ bd,
+                                    const float* __restrict__ freqs,
+                                    float cutoff, int pexp, int nrbf,
+                                    float* __restrict__ gbv_total, int64_t E) {
+    __shared__ float fsh[16];
+    if (threadIdx.x < nrbf) fsh[threadIdx.x] = freqs[threadIdx.x];
+    __syncthreads();
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < E; t += (int64_t)gridDim.x * blockDim.x) {
+        const float d = bd[t];
+        float gd = go_bd ? go_bd[t] : 0.0f;
+        gd += rbf_env_grad_dot<16>(d, fsh, cutoff, pexp, nrbf,
+                                   &go_exp[t * nrbf]);
+        const float scale = gd / d;
+        for (int k = 0; k < 3; ++k)
+            gbv_total[3 * t + k] = (go_bv ? go_bv[3 * t + k] : 0.0f)
+                + scale * bv[3 * t + k];
+    }
+}
+
+__global__ void k_rbf_env_fwd(const float* __restrict__ d,
+                              const float* __restrict__ freqs, float cutoff,
+                              int pexp, int nrbf, float* __restrict__ exp_out,
+                              int64_t M) {
+    __shared__ float fsh[16];
+    if (threadIdx.x < nrbf) fsh[threadIdx.x] = freqs[threadIdx.x];
+    __syncthreads();
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < M; t += (int64_t)gridDim.x * blockDim.x) {
+        float row[16];
+        rbf_env_row<16>(d[t], fsh, cutoff, pexp, nrbf, row);
+        for (int k = 0; k < nrbf; ++k) exp_out[t * nrbf + k] = row[k];
+    }
+}
+
+__global__ void k_rbf_env_bwd(const float* __restrict__ go_exp,
+                              const float* __restrict__ d,
+                              const float* __restrict__ freqs, float cutoff,
+                              int pexp, int nrbf, float* __restrict__ gd,
+                              int64_t M) {
+    __shared__ float fsh[16];
+    if (threadIdx.x < nrbf) fsh[threadIdx.x] = freqs[threadIdx.x];
+    __syncthreads();
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < M; t += (int64_t)gridDim.x * blockDim.x) {
+        gd[t] = rbf_env_grad_dot<16>(d[t], fsh, cutoff, pexp, nrbf,
+                                     &go_exp[t * nrbf]);
+    }
+}
+
 __device__ __forceinline__ float sigf(float x) {
-    return 1.0f / (1.0f + __expf(-x));
+    return 1.0f / (1.0f + expf(-x));
 }
 
 // out = base + silu(c) * sigmoid(g) * w   (gated-MLP epilogue, fused)
@@ -353,6 +491,54 @@ int dm_gated_combine_bwd_f32(const float* go, const float* c, const float* g,
     hipStream_t s = (hipStream_t)stream;
     k_gated_combine_bwd<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
         go, c, g, w, dc, dg, dw, total);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_edge_geom_rbf_fwd_f32(const float* pos, const int32_t* src,
+                             const int32_t* dst, const float* offshift,
+                             const float* freqs, float cutoff, int32_t pexp,
+                             int32_t nrbf, float* bv, float* bd,
+                             float* exp_out, int64_t E, uint64_t stream) {
+    if (nrbf > 16) { g_err = "nrbf > 16"; return -1; }
+    hipStream_t s = (hipStream_t)stream;
+    k_edge_geom_rbf_fwd<<<nblocks(E, BLOCK), BLOCK, 0, s>>>(
+        pos, src, dst, offshift, freqs, cutoff, pexp, nrbf, bv, bd, exp_out, E);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_edge_geom_rbf_bwd_f32(const float* go_bv, const float* go_bd,
+                             const float* go_exp, const float* bv,
+                             const float* bd, const float* freqs,
+                             float cutoff, int32_t pexp, int32_t nrbf,
+                             float* gbv_total, int64_t E, uint64_t stream) {
+    if (nrbf > 16) { g_err = "nrbf > 16"; return -1; }
+    hipStream_t s = (hipStream_t)stream;
+    k_edge_geom_rbf_bwd<<<nblocks(E, BLOCK), BLOCK, 0, s>>>(
+        go_bv, go_bd, go_exp, bv, bd, freqs, cutoff, pexp, nrbf, gbv_total, E);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_rbf_env_fwd_f32(const float* d, const float* freqs, float cutoff,
+                       int32_t pexp, int32_t nrbf, float* exp_out, int64_t M,
+                       uint64_t stream) {
+    if (nrbf > 16) { g_err = "nrbf > 16"; return -1; }
+    hipStream_t s = (hipStream_t)stream;
+    k_rbf_env_fwd<<<nblocks(M, BLOCK), BLOCK, 0, s>>>(
+        d, freqs, cutoff, pexp, nrbf, exp_out, M);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_rbf_env_bwd_f32(const float* go_exp, const float* d, const float* freqs,
+                       float cutoff, int32_t pexp, int32_t nrbf, float* gd,
+                       int64_t M, uint64_t stream) {
+    if (nrbf > 16) { g_err = "nrbf > 16"; return -1; }
+    hipStream_t s = (hipStream_t)stream;
+    k_rbf_env_bwd<<<nblocks(M, BLOCK), BLOCK, 0, s>>>(
+        go_exp, d, freqs, cutoff, pexp, nrbf, gd, M);
     DM_CHECK_LAUNCH();
     return 0;
 }

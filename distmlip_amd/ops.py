@@ -48,6 +48,21 @@ def hip_lib() -> ctypes.CDLL:
         lib.dm_gated_combine_bwd_f32.restype = c_int32
         lib.dm_gated_combine_bwd_f32.argtypes = [fp, fp, fp, fp, fp, fp, fp,
                                                  c_int64, c_uint64]
+        cf = ctypes.c_float
+        lib.dm_edge_geom_rbf_fwd_f32.restype = c_int32
+        lib.dm_edge_geom_rbf_fwd_f32.argtypes = [fp, ip, ip, fp, fp, cf,
+                                                 c_int32, c_int32, fp, fp, fp,
+                                                 c_int64, c_uint64]
+        lib.dm_edge_geom_rbf_bwd_f32.restype = c_int32
+        lib.dm_edge_geom_rbf_bwd_f32.argtypes = [fp, fp, fp, fp, fp, fp, cf,
+                                                 c_int32, c_int32, fp,
+                                                 c_int64, c_uint64]
+        lib.dm_rbf_env_fwd_f32.restype = c_int32
+        lib.dm_rbf_env_fwd_f32.argtypes = [fp, fp, cf, c_int32, c_int32, fp,
+                                           c_int64, c_uint64]
+        lib.dm_rbf_env_bwd_f32.restype = c_int32
+        lib.dm_rbf_env_bwd_f32.argtypes = [fp, fp, fp, cf, c_int32, c_int32,
+                                           fp, c_int64, c_uint64]
         lib.dm_hip_last_error.restype = c_char_p
         _hip_lib = lib
     return _hip_lib
@@ -229,6 +244,69 @@ class _GatedCombine(torch.autograd.Function):
         return dc, dg, dw, (go if ctx.has_base else None)
 
 
+class _EdgeGeomRbf(torch.autograd.Function):
+    """Fused bond_vec/bond_dist/RBF*envelope (chgnet.py:96-124 analog)."""
+
+    @staticmethod
+    def forward(ctx, pos, offshift, freqs, cutoff, pexp, pd):
+        _chk_f32(pos, offshift, freqs)
+        E = offshift.shape[0]
+        nrbf = freqs.shape[0]
+        bv = torch.empty(E, 3, dtype=pos.dtype, device=pos.device)
+        bd = torch.empty(E, dtype=pos.dtype, device=pos.device)
+        exp_out = torch.empty(E, nrbf, dtype=pos.dtype, device=pos.device)
+        _check(hip_lib().dm_edge_geom_rbf_fwd_f32(
+            _fp(pos), _ip(pd.src), _ip(pd.dst), _fp(offshift), _fp(freqs),
+            float(cutoff), int(pexp), nrbf, _fp(bv), _fp(bd), _fp(exp_out),
+            E, _stream()), "dm_edge_geom_rbf_fwd_f32")
+        ctx.save_for_backward(bv, bd, freqs)
+        ctx.pd = pd
+        ctx.cutoff, ctx.pexp, ctx.nrbf = float(cutoff), int(pexp), nrbf
+        ctx.n_nodes = pos.shape[0]
+        return bv, bd, exp_out
+
+    @staticmethod
+    def backward(ctx, go_bv, go_bd, go_exp):
+        bv, bd, freqs = ctx.saved_tensors
+        pd = ctx.pd
+        E = bv.shape[0]
+        gbv = torch.empty_like(bv)
+        go_exp = go_exp.contiguous() if go_exp is not None else \
+            torch.zeros(E, ctx.nrbf, dtype=bv.dtype, device=bv.device)
+        _check(hip_lib().dm_edge_geom_rbf_bwd_f32(
+            _fp(go_bv.contiguous()) if go_bv is not None else None,
+            _fp(go_bd.contiguous()) if go_bd is not None else None,
+            _fp(go_exp), _fp(bv), _fp(bd), _fp(freqs), ctx.cutoff, ctx.pexp,
+            ctx.nrbf, _fp(gbv), E, _stream()), "dm_edge_geom_rbf_bwd_f32")
+        g_pos = raw_seg_sum(gbv, pd.row_ptr, ctx.n_nodes) - \
+            raw_seg_sum_gather(gbv, pd.src_perm, pd.src_row_ptr, ctx.n_nodes)
+        # freqs gradient not implemented (inference engine; forces only)
+        return g_pos, gbv, None, None, None, None
+
+
+class _RbfEnv(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, d, freqs, cutoff, pexp):
+        _chk_f32(d, freqs)
+        M, nrbf = d.shape[0], freqs.shape[0]
+        out = torch.empty(M, nrbf, dtype=d.dtype, device=d.device)
+        _check(hip_lib().dm_rbf_env_fwd_f32(
+            _fp(d), _fp(freqs), float(cutoff), int(pexp), nrbf, _fp(out), M,
+            _stream()), "dm_rbf_env_fwd_f32")
+        ctx.save_for_backward(d, freqs)
+        ctx.cutoff, ctx.pexp, ctx.nrbf = float(cutoff), int(pexp), nrbf
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        d, freqs = ctx.saved_tensors
+        gd = torch.empty_like(d)
+        _check(hip_lib().dm_rbf_env_bwd_f32(
+            _fp(go.contiguous()), _fp(d), _fp(freqs), ctx.cutoff, ctx.pexp,
+            ctx.nrbf, _fp(gd), d.shape[0], _stream()), "dm_rbf_env_bwd_f32")
+        return gd, None, None, None
+
+
 class HipOps:
     """Product ops backend (see ops_base.OpsBackend)."""
 
@@ -260,3 +338,10 @@ class HipOps:
             c.contiguous(), g.contiguous(),
             w.contiguous() if w is not None else None,
             base.contiguous() if base is not None else None)
+
+    def edge_geom_rbf(self, pos, offshift, freqs, cutoff, pexp, pd):
+        return _EdgeGeomRbf.apply(pos.contiguous(), offshift.contiguous(),
+                                  freqs.contiguous(), cutoff, pexp, pd)
+
+    def rbf_env(self, d, freqs, cutoff, pexp):
+        return _RbfEnv.apply(d.contiguous(), freqs.contiguous(), cutoff, pexp)

@@ -227,14 +227,9 @@ class SpmdEngine:
                                      dtype=ft, device=dev)
         offshift = off_local @ lattice
 
-        bond_vec = ops.gather(pos, pd.dst, csr=(None, getattr(pd, "row_ptr", None))) \
-            + offshift - ops.gather(pos, pd.src,
-                                    csr=(getattr(pd, "src_perm", None),
-                                         getattr(pd, "src_row_ptr", None)))
-        bond_dist = torch.linalg.norm(bond_vec, dim=1)
-
-        bond_expansion = bond_expansion_from_dist(
-            bond_dist, core.rbf_freq_atom, cfg.cutoff, cfg.cutoff_exponent)
+        bond_vec, bond_dist, bond_expansion = ops.edge_geom_rbf(
+            pos, offshift, core.rbf_freq_atom, cfg.cutoff,
+            cfg.cutoff_exponent, pd)
 
         v = core.atom_embedding(species_local)
         e = core.bond_embedding(bond_expansion)
@@ -252,9 +247,8 @@ class SpmdEngine:
             nd_dist = HaloExchange.apply(nd_dist.unsqueeze(1), line_plan).squeeze(1)
             nd_vec = HaloExchange.apply(nd_vec, line_plan)
 
-            exp3 = bond_expansion_from_dist(
-                nd_dist, core.rbf_freq_bond, cfg.three_body_cutoff,
-                cfg.cutoff_exponent)
+            exp3 = ops.rbf_env(nd_dist, core.rbf_freq_bond,
+                               cfg.three_body_cutoff, cfg.cutoff_exponent)
             theta = compute_theta(
                 ops.gather(nd_vec, pd.l_src, csr=pd.line_src_csr),
                 ops.gather(nd_vec, pd.l_dst, csr=pd.line_dst_csr))
@@ -294,10 +288,14 @@ class SpmdEngine:
                 v = HaloExchange.apply(v, plan)
 
                 blk = core.bond_convs[layer_i]
-                msg = gated_mlp_split4(
-                    blk.bond_mlp, n, a, v, pd, ops, d,
-                    w=ops.gather(w_3b, pd.l_src, csr=pd.line_src_csr))
-                n = ops.scatter_lines(msg, pd, base=n)
+
+                def bond_body(n, a, v, w3, _blk=blk):
+                    msg = gated_mlp_split4(
+                        _blk.bond_mlp, n, a, v, pd, ops, d,
+                        w=ops.gather(w3, pd.l_src, csr=pd.line_src_csr))
+                    return ops.scatter_lines(msg, pd, base=n)
+
+                n = _ck(bond_body, n, a, v, w_3b)
                 e = e.index_copy(0, pd.map_de, n[pd.map_ude])    # bond_to_edge
 
                 if layer_i < cfg.n_blocks - 2:
@@ -306,8 +304,12 @@ class SpmdEngine:
                     # still computes it, chgnet.py:353-368) — skip it and
                     # the halo that feeds it
                     n = HaloExchange.apply(n, line_plan)
-                    a = gated_mlp_split4(blk.angle_mlp, n, a, v, pd, ops, d,
-                                         base=a)
+
+                    def angle_body(n, a, v, _blk=blk):
+                        return gated_mlp_split4(_blk.angle_mlp, n, a, v, pd,
+                                                ops, d, base=a)
+
+                    a = _ck(angle_body, n, a, v)
             else:
                 v = HaloExchange.apply(v, plan)
 

@@ -70,6 +70,36 @@ int dm_gated_combine_bwd_f32(const float* go, const float* c, const float* g,
                              const float* w, float* dc, float* dg, float* dw,
                              int64_t total, uint64_t stream);
 
+/* Fused edge geometry + radial basis (replaces the torch chain for
+ * chgnet.py:96-124): per local edge e,
+ *   bv[e]  = pos[dst[e]] + offshift[e] - pos[src[e]]
+ *   bd[e]  = |bv[e]|
+ *   exp[e,k] = env(rbf_k) * rbf_k,   rbf_k = sqrt(2/c) sin(f_k bd / c)/bd
+ * env = the reference's polynomial cutoff applied to the RBF VALUE
+ * (chgnet.py:119-121 quirk), exponent `pexp`.  nrbf <= 16. */
+int dm_edge_geom_rbf_fwd_f32(const float* pos, const int32_t* src,
+                             const int32_t* dst, const float* offshift,
+                             const float* freqs, float cutoff, int32_t pexp,
+                             int32_t nrbf, float* bv, float* bd, float* exp_out,
+                             int64_t E, uint64_t stream);
+/* backward: gbv_total[e] = go_bv[e] + (go_bd[e] + sum_k go_exp[e,k] *
+ * d(exp_k)/d(bd)) * bv[e]/bd[e].  The caller scatters gbv_total to pos via
+ * the CSRs and passes it on as d(offshift).  No gradient wrt freqs. */
+int dm_edge_geom_rbf_bwd_f32(const float* go_bv, const float* go_bd,
+                             const float* go_exp, const float* bv,
+                             const float* bd, const float* freqs,
+                             float cutoff, int32_t pexp, int32_t nrbf,
+                             float* gbv_total, int64_t E, uint64_t stream);
+
+/* Fused radial-basis + envelope only (bond-graph expansion on nd_dist):
+ * exp[m,k] = env(rbf_k(d[m])) * rbf_k(d[m]); bwd gives gd[m]. */
+int dm_rbf_env_fwd_f32(const float* d, const float* freqs, float cutoff,
+                       int32_t pexp, int32_t nrbf, float* exp_out, int64_t M,
+                       uint64_t stream);
+int dm_rbf_env_bwd_f32(const float* go_exp, const float* d, const float* freqs,
+                       float cutoff, int32_t pexp, int32_t nrbf, float* gd,
+                       int64_t M, uint64_t stream);
+
 const char* dm_hip_last_error(void);
 
 #ifdef __cplusplus

@@ -215,3 +215,13 @@ class CpuRefOps:
         if w is not None:
             out = out * w
         return out if base is None else base + out
+
+    def edge_geom_rbf(self, pos, offshift, freqs, cutoff, pexp, pd):
+        from distmlip_amd.model import bond_expansion_from_dist
+        bv = pos[pd.dst] + offshift - pos[pd.src]
+        bd = torch.linalg.norm(bv, dim=1)
+        return bv, bd, bond_expansion_from_dist(bd, freqs, cutoff, pexp)
+
+    def rbf_env(self, d, freqs, cutoff, pexp):
+        from distmlip_amd.model import bond_expansion_from_dist
+        return bond_expansion_from_dist(d, freqs, cutoff, pexp)
