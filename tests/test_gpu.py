@@ -87,6 +87,60 @@ class TestKernels:
 
 
 @requires_gpu
+def test_edge_geom_rbf_fwd_bwd():
+    """Fused geometry+RBF kernel vs the torch composition (fwd + grads)."""
+    import math
+
+    from distmlip_amd.model import bond_expansion_from_dist
+    from distmlip_amd.ops import _EdgeGeomRbf
+    torch.manual_seed(3)
+    dev = torch.device("cuda:0")
+    N, E = 200, 3000
+
+    class PD:
+        pass
+    pd = PD()
+    src64 = torch.randint(0, N, (E,))
+    dst64 = torch.sort(torch.randint(0, N, (E,))).values
+    pd.src = src64.to(torch.int32).to(dev)
+    pd.dst = dst64.to(torch.int32).to(dev)
+
+    def csr(idx):
+        import numpy as np
+        idx = idx.numpy()
+        order = np.argsort(idx, kind="stable")
+        rp = np.zeros(N + 1, dtype=np.int64)
+        np.add.at(rp, idx + 1, 1)
+        return (torch.tensor(order, dtype=torch.int32, device=dev),
+                torch.tensor(np.cumsum(rp), dtype=torch.int32, device=dev))
+    pd.src_perm, pd.src_row_ptr = csr(src64)
+    _, pd.row_ptr = csr(dst64)
+
+    pos = (torch.rand(N, 3, device=dev) * 20).requires_grad_(True)
+    off = (torch.randn(E, 3, device=dev) * 2).requires_grad_(True)
+    freqs = (torch.arange(1, 10, device=dev, dtype=torch.float32) * math.pi)
+    bv, bd, exp = _EdgeGeomRbf.apply(pos, off, freqs, 6.0, 5, pd)
+
+    pos2 = pos.detach().clone().requires_grad_(True)
+    off2 = off.detach().clone().requires_grad_(True)
+    bv2 = pos2[dst64.to(dev)] + off2 - pos2[src64.to(dev)]
+    bd2 = torch.linalg.norm(bv2, dim=1)
+    exp2 = bond_expansion_from_dist(bd2, freqs, 6.0, 5)
+    assert torch.allclose(bv, bv2, atol=1e-5)
+    assert torch.allclose(bd, bd2, atol=1e-5)
+    assert torch.allclose(exp, exp2, atol=1e-5), (exp - exp2).abs().max()
+
+    g_bv = torch.randn_like(bv)
+    g_bd = torch.randn_like(bd)
+    g_exp = torch.randn_like(exp)
+    torch.autograd.backward([bv, bd, exp], [g_bv, g_bd, g_exp])
+    torch.autograd.backward([bv2, bd2, exp2], [g_bv, g_bd, g_exp])
+    assert torch.allclose(pos.grad, pos2.grad, atol=2e-4, rtol=1e-3), \
+        (pos.grad - pos2.grad).abs().max()
+    assert torch.allclose(off.grad, off2.grad, atol=2e-4, rtol=1e-3)
+
+
+@requires_gpu
 def test_gated_combine_fwd_bwd():
     from distmlip_amd.ops import _GatedCombine
     torch.manual_seed(1)
