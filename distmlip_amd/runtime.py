@@ -152,6 +152,10 @@ class SpmdEngine:
             device = f"cuda:{torch.cuda.current_device()}"
         self.device = torch.device(device)
         self.core = deepcopy(core).to(self.device).eval()
+        # inference engine: frozen parameters mean no GEMM saves its
+        # [E,*] input for dW and backward drops every dW GEMM — large
+        # memory and time win on the force backward
+        self.core.requires_grad_(False)
         self.ops = ops if ops is not None else default_ops_factory(self.device)
         self.threads = threads
         self.graph_backend = graph_backend
@@ -233,8 +237,6 @@ class SpmdEngine:
 
         v = core.atom_embedding(species_local)
         e = core.bond_embedding(bond_expansion)
-        w_ab = core.atom_bond_weights(bond_expansion)
-        w_bb = core.bond_bond_weights(bond_expansion)
 
         use_bg = self.use_bond_graph
         if use_bg:
@@ -253,7 +255,6 @@ class SpmdEngine:
                 ops.gather(nd_vec, pd.l_src, csr=pd.line_src_csr),
                 ops.gather(nd_vec, pd.l_dst, csr=pd.line_dst_csr))
             a = core.angle_embedding(fourier_expansion(theta, core.angle_freq))
-            w_3b = core.threebody_bond_weights(exp3)
             n = torch.zeros(pd.n_bonds, cfg.dim, dtype=ft, device=dev).index_copy(
                 0, pd.map_ude, e[pd.map_de])
             n = HaloExchange.apply(n, line_plan)
@@ -268,17 +269,32 @@ class SpmdEngine:
                     fn, *args, use_reentrant=False)
             return fn(*args)
 
-        def atom_conv_body(layer_idx_t, v, e):
+        # shared message weights (chgnet.py:272-294).  Under checkpointing
+        # the [E,64]/[B,64] weight tensors are recomputed inside each
+        # segment from the small [*,9] expansions instead of being held
+        # for the whole forward (2x [E,64] = 22 GB at 1M atoms).
+        if ckpt:
+            w_ab = w_bb = None
+            w_3b = None
+        else:
+            w_ab = core.atom_bond_weights(bond_expansion)
+            w_bb = core.bond_bond_weights(bond_expansion)
+            w_3b = core.threebody_bond_weights(exp3) if use_bg else None
+
+        def atom_conv_body(layer_idx_t, v, e, bexp):
             blk = core.atom_convs[int(layer_idx_t)]
+            wbb = w_bb if w_bb is not None else core.bond_bond_weights(bexp)
+            wab = w_ab if w_ab is not None else core.atom_bond_weights(bexp)
             e = gated_mlp_split3(blk.edge_mlp, v, e, pd, ops, d,
-                                 w=w_bb, base=e)
-            msg = gated_mlp_split3(blk.node_mlp, v, e, pd, ops, d, w=w_ab)
+                                 w=wbb, base=e)
+            msg = gated_mlp_split3(blk.node_mlp, v, e, pd, ops, d, w=wab)
             v = ops.scatter_edges(msg, pd, base=v)
             return v, e
 
         def atom_conv(layer_i, v, e):
             return _ck(atom_conv_body,
-                       torch.tensor(layer_i % cfg.n_blocks), v, e)
+                       torch.tensor(layer_i % cfg.n_blocks), v, e,
+                       bond_expansion)
 
         for layer_i in range(cfg.n_blocks - 1):           # chgnet.py:296-368
             v, e = atom_conv(layer_i, v, e)
@@ -289,13 +305,15 @@ class SpmdEngine:
 
                 blk = core.bond_convs[layer_i]
 
-                def bond_body(n, a, v, w3, _blk=blk):
+                def bond_body(n, a, v, e3, _blk=blk):
+                    w3 = w_3b if w_3b is not None else \
+                        core.threebody_bond_weights(e3)
                     msg = gated_mlp_split4(
                         _blk.bond_mlp, n, a, v, pd, ops, d,
                         w=ops.gather(w3, pd.l_src, csr=pd.line_src_csr))
                     return ops.scatter_lines(msg, pd, base=n)
 
-                n = _ck(bond_body, n, a, v, w_3b)
+                n = _ck(bond_body, n, a, v, exp3)
                 e = e.index_copy(0, pd.map_de, n[pd.map_ude])    # bond_to_edge
 
                 if layer_i < cfg.n_blocks - 2:
