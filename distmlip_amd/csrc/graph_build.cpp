@@ -36,6 +36,13 @@
 #include <omp.h>
 
 #include <malloc.h>
+#include <cstdlib>
+
+#define DM_T(name) \
+    do { if (getenv("DM_TIMING")) { \
+        std::fprintf(stderr, "DM_TIMING %-22s %.1f ms\n", name, \
+                     (omp_get_wtime() - t_phase_) * 1e3); \
+        t_phase_ = omp_get_wtime(); } } while (0)
 
 namespace {
 
@@ -82,19 +89,67 @@ struct Part {
 };
 
 // stable counting sort of [0..m) by key[i] in [0..nkeys); emits perm and
-// CSR row_ptr[nkeys+1]
+// CSR row_ptr[nkeys+1].  Parallel two-pass with per-thread count matrices
+// for large m (thread-count-independent output: chunk t's elements of
+// bucket k land at base[k] + sum of earlier chunks' counts — stable).
 template <class K>
 void counting_csr(const K* key, int64_t m, int64_t nkeys,
                   rvec<int32_t>& perm, rvec<int32_t>& rptr) {
-    std::vector<int64_t> cnt(nkeys + 1, 0);
-    for (int64_t i = 0; i < m; ++i) ++cnt[key[i] + 1];
-    for (int64_t k = 0; k < nkeys; ++k) cnt[k + 1] += cnt[k];
-    rptr.resize(nkeys + 1);
-    for (int64_t k = 0; k <= nkeys; ++k) rptr[k] = (int32_t)cnt[k];
     perm.resize(m);
-    std::vector<int64_t> fill(nkeys, 0);
-    for (int64_t i = 0; i < m; ++i)
-        perm[cnt[key[i]] + fill[key[i]]++] = (int32_t)i;
+    rptr.resize(nkeys + 1);
+    const int T = (m > 2'000'000) ? std::min(omp_get_max_threads(), 16) : 1;
+    if (T == 1) {
+        std::vector<int64_t> cnt(nkeys + 1, 0);
+        for (int64_t i = 0; i < m; ++i) ++cnt[key[i] + 1];
+        for (int64_t k = 0; k < nkeys; ++k) cnt[k + 1] += cnt[k];
+        for (int64_t k = 0; k <= nkeys; ++k) rptr[k] = (int32_t)cnt[k];
+        std::vector<int64_t> fill(nkeys, 0);
+        for (int64_t i = 0; i < m; ++i)
+            perm[cnt[key[i]] + fill[key[i]]++] = (int32_t)i;
+        return;
+    }
+    rvec<int64_t> tcnt;
+    tcnt.resize((int64_t)T * nkeys);
+#pragma omp parallel num_threads(T)
+    {
+        const int t = omp_get_thread_num();
+        int64_t* c = &tcnt[(int64_t)t * nkeys];
+        std::memset(c, 0, nkeys * sizeof(int64_t));
+        const int64_t lo = m * t / T, hi = m * (t + 1) / T;
+        for (int64_t i = lo; i < hi; ++i) ++c[key[i]];
+    }
+    // exclusive scan over (bucket-major, thread-minor): c[t][k] becomes the
+    // start offset for chunk t's elements of bucket k.  Blocked so the
+    // strided (t, k) accesses stay cache-resident.
+    int64_t run = 0;
+    rptr[0] = 0;
+    constexpr int64_t BL = 8192;
+    std::vector<int64_t> blk((int64_t)T * BL);
+    for (int64_t kb = 0; kb < nkeys; kb += BL) {
+        const int64_t be = std::min(nkeys, kb + BL), bw = be - kb;
+        for (int t = 0; t < T; ++t)
+            std::memcpy(&blk[t * bw], &tcnt[(int64_t)t * nkeys + kb],
+                        bw * sizeof(int64_t));
+        for (int64_t k = 0; k < bw; ++k) {
+            for (int t = 0; t < T; ++t) {
+                const int64_t v = blk[t * bw + k];
+                blk[t * bw + k] = run;
+                run += v;
+            }
+            rptr[kb + k + 1] = (int32_t)run;
+        }
+        for (int t = 0; t < T; ++t)
+            std::memcpy(&tcnt[(int64_t)t * nkeys + kb], &blk[t * bw],
+                        bw * sizeof(int64_t));
+    }
+#pragma omp parallel num_threads(T)
+    {
+        const int t = omp_get_thread_num();
+        int64_t* c = &tcnt[(int64_t)t * nkeys];
+        const int64_t lo = m * t / T, hi = m * (t + 1) / T;
+        for (int64_t i = lo; i < hi; ++i)
+            perm[c[key[i]]++] = (int32_t)i;
+    }
 }
 
 }  // namespace
@@ -116,6 +171,7 @@ int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
                         const int64_t* pbc, int64_t n, double r, double bond_r,
                         double tol, int nthreads) {
     const double r2 = r * r, bond_r2 = bond_r * bond_r;
+    double t_phase_ = omp_get_wtime();
 
     std::vector<double> cart(3 * n);
 #pragma omp parallel for num_threads(nthreads) schedule(static)
@@ -203,6 +259,7 @@ int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
         }
     }
 
+    DM_T("nl:expand");
     // linked-cell grid over the padded bbox, cell size >= r
     const int64_t nx = std::max<int64_t>(1, (int64_t)((vmax[0] - vmin[0]) / r));
     const int64_t ny = std::max<int64_t>(1, (int64_t)((vmax[1] - vmin[1]) / r));
@@ -228,6 +285,7 @@ int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
     for (int64_t j = 0; j < nxp; ++j)
         cell_pts[cell_cnt[pt_cell[j]] + cell_fill[pt_cell[j]]++] = (int32_t)j;
 
+    DM_T("nl:bin");
     // per-center count + fill (deterministic center-major order)
     std::vector<int64_t> ecnt(n + 1, 0), wcnt(n + 1, 0);
     auto visit = [&](int64_t i, auto&& emit) {
@@ -254,6 +312,7 @@ int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
         visit(i, [&](const XP&, double d2) { ++e; if (d2 < bond_r2 + tol) ++w; });
         ecnt[i + 1] = e; wcnt[i + 1] = w;
     }
+    DM_T("nl:count");
     for (int64_t i = 0; i < n; ++i) { ecnt[i + 1] += ecnt[i]; wcnt[i + 1] += wcnt[i]; }
 
     const int64_t E = ecnt[n], W = wcnt[n];
@@ -271,6 +330,7 @@ int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
             ++e;
         });
     }
+    DM_T("nl:fill");
     return 0;
 }
 
@@ -292,6 +352,7 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
                      double r, double bond_r, int P, int nthreads,
                      bool use_bond_graph) {
     const int64_t n = g->n_atoms, E = g->n_edges;
+    double t_phase_ = omp_get_wtime();
     g->P = P;
     g->parts.resize(P);
 
@@ -415,6 +476,7 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
 #pragma omp parallel for num_threads(nthreads) schedule(static)
     for (int64_t e = 0; e < E; ++e) eowner[e] = home[g->dst[e]];   // utils.c:206
 
+    DM_T("pt:classify");
     for (int p = 0; p < P; ++p) {
         Part& pt = g->parts[p];
         std::fill(g2l_node.begin(), g2l_node.end(), (int64_t)-1);
@@ -470,6 +532,7 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
             }
             counting_csr(pt.src_local.data(), Ep, Nn, pt.src_perm, pt.src_row_ptr);
         }
+        DM_T("pt:edges");
 
         if (!use_bond_graph) continue;
 
@@ -572,6 +635,7 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
                      pt.line_src_row_ptr);
         counting_csr(pt.line_center.data(), L, (int64_t)pt.global_ids.size(),
                      pt.center_perm, pt.center_row_ptr);
+        DM_T("pt:bonds+lines");
     }
     return 0;
 }
