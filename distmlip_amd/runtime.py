@@ -241,10 +241,12 @@ class SpmdEngine:
         use_bg = self.use_bond_graph
         if use_bg:
             # owned bond geometry from local edges; ghosts by one halo each
-            # (reference chgnet.py:129-164)
-            nd_dist = torch.zeros(pd.n_bonds, dtype=ft, device=dev).index_copy(
+            # (reference chgnet.py:129-164).  map_ude is a permutation of
+            # the owned range and the halo fills every ghost slot, so the
+            # buffers start uninitialized (no zero-fill pass).
+            nd_dist = torch.empty(pd.n_bonds, dtype=ft, device=dev).index_copy(
                 0, pd.map_ude, bond_dist[pd.map_de])
-            nd_vec = torch.zeros(pd.n_bonds, 3, dtype=ft, device=dev).index_copy(
+            nd_vec = torch.empty(pd.n_bonds, 3, dtype=ft, device=dev).index_copy(
                 0, pd.map_ude, bond_vec[pd.map_de])
             nd_dist = HaloExchange.apply(nd_dist.unsqueeze(1), line_plan).squeeze(1)
             nd_vec = HaloExchange.apply(nd_vec, line_plan)
@@ -255,7 +257,7 @@ class SpmdEngine:
                 ops.gather(nd_vec, pd.l_src, csr=pd.line_src_csr),
                 ops.gather(nd_vec, pd.l_dst, csr=pd.line_dst_csr))
             a = core.angle_embedding(fourier_expansion(theta, core.angle_freq))
-            n = torch.zeros(pd.n_bonds, cfg.dim, dtype=ft, device=dev).index_copy(
+            n = torch.empty(pd.n_bonds, cfg.dim, dtype=ft, device=dev).index_copy(
                 0, pd.map_ude, e[pd.map_de])
             n = HaloExchange.apply(n, line_plan)
 

@@ -11,7 +11,7 @@ import torch.distributed as dist
 import torch.multiprocessing as mp
 
 
-def _worker(rank, world, init_file, out_dir, calc_stresses):
+def _worker(rank, world, init_file, out_dir, calc_stresses, checkpoint="auto"):
     import sys
     sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     from distmlip_amd.model import CHGNetCore
@@ -24,7 +24,8 @@ def _worker(rank, world, init_file, out_dir, calc_stresses):
     try:
         s = diamond_si((12, 2, 2), jitter=0.12, seed=2)
         core = CHGNetCore.seeded(seed=0).double()
-        eng = SpmdEngine(core, world, threads=2, device="cpu", ops=CpuRefOps())
+        eng = SpmdEngine(core, world, threads=2, device="cpu", ops=CpuRefOps(),
+                         checkpoint=checkpoint)
         out = eng.step(s, calc_stresses=calc_stresses)
         np.save(os.path.join(out_dir, f"E_{rank}.npy"),
                 np.array([out["energy"].item()]))
@@ -39,17 +40,22 @@ def _worker(rank, world, init_file, out_dir, calc_stresses):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("calc_stresses", [False, True])
-def test_spmd_two_ranks_matches_oracle(calc_stresses, tmp_path):
+@pytest.mark.parametrize("world,calc_stresses,checkpoint", [
+    (2, False, "auto"),
+    (2, True, "auto"),
+    (3, False, "auto"),
+    (2, False, "on"),     # forced activation checkpointing (1M-atom path)
+])
+def test_spmd_ranks_match_oracle(world, calc_stresses, checkpoint, tmp_path):
     from distmlip_amd.model import CHGNetCore
     from distmlip_amd.structures import diamond_si
     from oracle.chgnet_ref import oracle_forward
     from oracle.graph_ref import brute_force_neighbors
 
-    world = 2
     init_file = str(tmp_path / "pg_init")
     out_dir = str(tmp_path)
-    mp.spawn(_worker, args=(world, init_file, out_dir, calc_stresses),
+    mp.spawn(_worker, args=(world, init_file, out_dir, calc_stresses,
+                            checkpoint),
              nprocs=world, join=True)
 
     s = diamond_si((12, 2, 2), jitter=0.12, seed=2)
