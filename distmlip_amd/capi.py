@@ -44,6 +44,11 @@ def graph_lib() -> ctypes.CDLL:
             POINTER(c_double), POINTER(c_double), POINTER(c_int64), c_int64,
             c_double, c_double, c_double, c_int32, c_int32, c_int32,
             POINTER(c_void_p)]
+        lib.dm_graph_build_focus.restype = c_int32
+        lib.dm_graph_build_focus.argtypes = [
+            POINTER(c_double), POINTER(c_double), POINTER(c_int64), c_int64,
+            c_double, c_double, c_double, c_int32, c_int32, c_int32, c_int32,
+            POINTER(c_void_p)]
         lib.dm_graph_global_view.restype = c_int32
         lib.dm_graph_partition_view.restype = c_int32
         lib.dm_graph_free.argtypes = [c_void_p]
@@ -114,21 +119,22 @@ def _as_np(ptr, count, nptype, owner, shape=None):
 
 
 def build_graph(frac_coords, lattice, pbc, cutoff, bond_cutoff, tol,
-                num_partitions, num_threads, use_bond_graph):
-    """Low-level build; returns (owner, global dict, [partition dicts])."""
+                num_partitions, num_threads, use_bond_graph, focus=-1):
+    """Low-level build; returns (owner, global dict, [partition dicts]).
+    focus >= 0 = the SPMD per-rank slab build (see dm_graph_build_focus)."""
     lib = graph_lib()
     frac = np.ascontiguousarray(frac_coords, dtype=np.float64)
     lat = np.ascontiguousarray(lattice, dtype=np.float64)
     pbc = np.ascontiguousarray(pbc, dtype=np.int64)
     n = len(frac)
     h = c_void_p()
-    rc = lib.dm_graph_build(
+    rc = lib.dm_graph_build_focus(
         frac.ctypes.data_as(POINTER(c_double)),
         lat.ctypes.data_as(POINTER(c_double)),
         pbc.ctypes.data_as(POINTER(c_int64)),
         c_int64(n), c_double(cutoff), c_double(bond_cutoff), c_double(tol),
         c_int32(num_partitions), c_int32(num_threads),
-        c_int32(1 if use_bond_graph else 0), ctypes.byref(h))
+        c_int32(1 if use_bond_graph else 0), c_int32(focus), ctypes.byref(h))
     if rc != 0:
         msg = lib.dm_last_error().decode()
         if rc == -4:
@@ -185,14 +191,14 @@ def build_graph(frac_coords, lattice, pbc, cutoff, bond_cutoff, tol,
 
 def get_subgraphs_fast(cart_coords, cutoff, pbc, lattice, num_partitions,
                        bond_cutoff, tol, num_threads, use_bond_graph,
-                       frac_coords, return_csr=False):
+                       frac_coords, return_csr=False, focus=-1):
     """Reference-compatible entry (subgraph_creation_fast.c:92-453 tuple
     + one extension element: per-partition per-BDE global edge ids).
     With return_csr=True also returns the per-partition CSR dicts the HIP
     kernel path consumes."""
     _owner, g, parts = build_graph(frac_coords, lattice, pbc, cutoff,
                                    bond_cutoff, tol, num_partitions,
-                                   num_threads, use_bond_graph)
+                                   num_threads, use_bond_graph, focus=focus)
     frac = np.ascontiguousarray(frac_coords, dtype=np.float64)
     lat = np.ascontiguousarray(lattice, dtype=np.float64)
     wrapped_cart = frac @ lat

@@ -166,10 +166,16 @@ namespace {
 
 // ---------------------------------------------------------------------------
 // neighbor list (contract of fpis.c:418-901)
+//
+// `frac` holds n PACKED (possibly subset) atoms; `sel` (nullable) maps the
+// packed index to the GLOBAL atom id.  Edges are emitted center-major with
+// each center's neighbors in CANONICAL (global dst id, image) order, so a
+// focused slab build and a full build produce identical relative orders —
+// the property the cross-rank halo slice alignment relies on.
 // ---------------------------------------------------------------------------
 int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
                         const int64_t* pbc, int64_t n, double r, double bond_r,
-                        double tol, int nthreads) {
+                        double tol, int nthreads, const int64_t* sel) {
     const double r2 = r * r, bond_r2 = bond_r * bond_r;
     double t_phase_ = omp_get_wtime();
 
@@ -319,16 +325,34 @@ int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
     g->n_edges = E; g->n_within = W;
     g->src.resize(E); g->dst.resize(E); g->offsets.resize(3 * E); g->dist.resize(E);
     g->within.resize(W);
-#pragma omp parallel for num_threads(nthreads) schedule(dynamic, 256)
-    for (int64_t i = 0; i < n; ++i) {
-        int64_t e = ecnt[i], w = wcnt[i];
-        visit(i, [&](const XP& p, double d2) {
-            g->src[e] = i; g->dst[e] = p.orig;
-            g->offsets[3 * e] = p.ia; g->offsets[3 * e + 1] = p.ib; g->offsets[3 * e + 2] = p.ic;
-            g->dist[e] = std::sqrt(d2);
-            if (d2 < bond_r2 + tol) g->within[w++] = e;
-            ++e;
-        });
+    struct ERec { int64_t dst, key; int16_t ia, ib, ic; double d2; };
+#pragma omp parallel num_threads(nthreads)
+    {
+        std::vector<ERec> buf;
+        buf.reserve(512);
+#pragma omp for schedule(dynamic, 256)
+        for (int64_t i = 0; i < n; ++i) {
+            buf.clear();
+            visit(i, [&](const XP& p, double d2) {
+                buf.push_back(ERec{p.orig, sel ? sel[p.orig] : p.orig,
+                                   p.ia, p.ib, p.ic, d2});
+            });
+            std::sort(buf.begin(), buf.end(), [](const ERec& a, const ERec& b) {
+                if (a.key != b.key) return a.key < b.key;
+                if (a.ia != b.ia) return a.ia < b.ia;
+                if (a.ib != b.ib) return a.ib < b.ib;
+                return a.ic < b.ic;
+            });
+            int64_t e = ecnt[i], w = wcnt[i];
+            for (const ERec& p : buf) {
+                g->src[e] = i; g->dst[e] = p.dst;
+                g->offsets[3 * e] = p.ia; g->offsets[3 * e + 1] = p.ib;
+                g->offsets[3 * e + 2] = p.ic;
+                g->dist[e] = std::sqrt(p.d2);
+                if (p.d2 < bond_r2 + tol) g->within[w++] = e;
+                ++e;
+            }
+        }
     }
     DM_T("nl:fill");
     return 0;
@@ -339,6 +363,7 @@ int build_neighbor_list(dm_graph* g, const double* frac, const double* lat,
 // ---------------------------------------------------------------------------
 struct Walls {
     int dim = 0;
+    double fmin = 0, fmax = 0;
     std::vector<double> w;
     inline int which(const double* fc) const {                  // utils.c:1312-1322
         const double x = fc[dim];
@@ -348,70 +373,77 @@ struct Walls {
     }
 };
 
-int build_partitions(dm_graph* g, const double* frac, const double* lat,
+// partition rule over the FULL structure (utils.c:1370-1456 + width check
+// utils.c:1512-1529).  Always global: a focused slab build must use the
+// same walls as every other rank.
+int compute_walls(const double* frac, const double* lat, int64_t n, int P,
+                  double r, double bond_r, bool use_bond_graph, Walls& walls) {
+    if (P <= 1) return 0;
+    // QUIRK preserved: the reference picks the partition dimension from
+    // "cartesian" coordinates computed as cart = L . frac with lattice
+    // ROWS as the matrix rows (fast.c:16-27 fractional_to_cartesian),
+    // i.e. the TRANSPOSE of the convention FPIS/ase use (cart = frac @ L).
+    double cmin[3] = {1e300, 1e300, 1e300}, cmax[3] = {-1e300, -1e300, -1e300};
+    for (int64_t i = 0; i < n; ++i) {
+        const double u = frac[3 * i], v = frac[3 * i + 1], w = frac[3 * i + 2];
+        double c[3];
+        for (int k = 0; k < 3; ++k)
+            c[k] = u * lat[3 * k] + v * lat[3 * k + 1] + w * lat[3 * k + 2];
+        for (int k = 0; k < 3; ++k) { cmin[k] = std::min(cmin[k], c[k]); cmax[k] = std::max(cmax[k], c[k]); }
+    }
+    int dim = 0;
+    for (int k = 1; k < 3; ++k)
+        if (cmax[k] - cmin[k] > cmax[dim] - cmin[dim]) dim = k;
+    double fmin = frac[dim], fmax = frac[dim];
+    for (int64_t i = 1; i < n; ++i) {
+        fmin = std::min(fmin, frac[3 * i + dim]);
+        fmax = std::max(fmax, frac[3 * i + dim]);
+    }
+    const double flen = fmax - fmin;
+    walls.dim = dim;
+    walls.fmin = fmin;
+    walls.fmax = fmax;
+    walls.w.resize(P - 1);
+    for (int i = 1; i < P; ++i)
+        walls.w[i - 1] = i * (flen / P) + WALL_EPSILON + fmin;   // utils.c:1433-1435
+    bool coll = true;                                            // utils.c:1440-1455
+    while (coll) {
+        coll = false;
+        for (int wi = 0; wi < P - 1; ++wi)
+            for (int64_t i = 0; i < n; ++i)
+                if (frac[3 * i + dim] == walls.w[wi]) {
+                    coll = true;
+                    std::fprintf(stderr, "Collision b/w atom and partition wall, moving wall.\n");
+                    walls.w[wi] += WALL_EPSILON;
+                }
+    }
+    const double lv[3] = {lat[dim], lat[dim + 3], lat[dim + 6]};
+    const double vnorm = std::sqrt(lv[0] * lv[0] + lv[1] * lv[1] + lv[2] * lv[2]);
+    const double width = walls.w[0] * vnorm;
+    if (use_bond_graph && width <= 2 * (r + bond_r)) {
+        set_err("Partition walls are too close together: width " +
+                std::to_string(width) + " <= 2*(cutoff+bond_cutoff)");
+        return -4;
+    }
+    if (!use_bond_graph && width <= 2 * r) {
+        set_err("Partition walls are too close together: width " +
+                std::to_string(width) + " <= 2*cutoff");
+        return -4;
+    }
+    return 0;
+}
+
+// `frac` holds n packed (possibly subset) atoms; `sel` maps packed -> global
+// atom id (nullptr = identity).  With focus >= 0 only that partition's
+// outputs are materialized (the others stay empty).
+int build_partitions(dm_graph* g, const double* frac, const int64_t* sel,
+                     int64_t n, const Walls& walls,
                      double r, double bond_r, int P, int nthreads,
-                     bool use_bond_graph) {
-    const int64_t n = g->n_atoms, E = g->n_edges;
+                     bool use_bond_graph, int focus) {
+    const int64_t E = g->n_edges;
     double t_phase_ = omp_get_wtime();
     g->P = P;
     g->parts.resize(P);
-
-    // --- partition rule (utils.c:1370-1456)
-    Walls walls;
-    if (P > 1) {
-        // QUIRK preserved: the reference picks the partition dimension from
-        // "cartesian" coordinates computed as cart = L . frac with lattice
-        // ROWS as the matrix rows (fast.c:16-27 fractional_to_cartesian),
-        // i.e. the TRANSPOSE of the convention FPIS/ase use (cart = frac @ L).
-        // Identical for diagonal lattices, different for skewed cells.
-        double cmin[3] = {1e300, 1e300, 1e300}, cmax[3] = {-1e300, -1e300, -1e300};
-        for (int64_t i = 0; i < n; ++i) {
-            const double u = frac[3 * i], v = frac[3 * i + 1], w = frac[3 * i + 2];
-            double c[3];
-            for (int k = 0; k < 3; ++k)
-                c[k] = u * lat[3 * k] + v * lat[3 * k + 1] + w * lat[3 * k + 2];
-            for (int k = 0; k < 3; ++k) { cmin[k] = std::min(cmin[k], c[k]); cmax[k] = std::max(cmax[k], c[k]); }
-        }
-        int dim = 0;
-        for (int k = 1; k < 3; ++k)
-            if (cmax[k] - cmin[k] > cmax[dim] - cmin[dim]) dim = k;
-        double fmin = frac[dim], fmax = frac[dim];
-        for (int64_t i = 1; i < n; ++i) {
-            fmin = std::min(fmin, frac[3 * i + dim]);
-            fmax = std::max(fmax, frac[3 * i + dim]);
-        }
-        const double flen = fmax - fmin;
-        walls.dim = dim;
-        walls.w.resize(P - 1);
-        for (int i = 1; i < P; ++i)
-            walls.w[i - 1] = i * (flen / P) + WALL_EPSILON + fmin;   // utils.c:1433-1435
-        bool coll = true;                                            // utils.c:1440-1455
-        while (coll) {
-            coll = false;
-            for (int wi = 0; wi < P - 1; ++wi)
-                for (int64_t i = 0; i < n; ++i)
-                    if (frac[3 * i + dim] == walls.w[wi]) {
-                        coll = true;
-                        std::fprintf(stderr, "Collision b/w atom and partition wall, moving wall.\n");
-                        walls.w[wi] += WALL_EPSILON;
-                    }
-        }
-        // width check (utils.c:1512-1529, quirks preserved: walls[0] and
-        // lattice COLUMN norm)
-        const double lv[3] = {lat[dim], lat[dim + 3], lat[dim + 6]};
-        const double vnorm = std::sqrt(lv[0] * lv[0] + lv[1] * lv[1] + lv[2] * lv[2]);
-        const double width = walls.w[0] * vnorm;
-        if (use_bond_graph && width <= 2 * (r + bond_r)) {
-            set_err("Partition walls are too close together: width " +
-                    std::to_string(width) + " <= 2*(cutoff+bond_cutoff)");
-            return -4;
-        }
-        if (!use_bond_graph && width <= 2 * r) {
-            set_err("Partition walls are too close together: width " +
-                    std::to_string(width) + " <= 2*cutoff");
-            return -4;
-        }
-    }
 
     // home partition per node
     std::vector<int32_t> home(n);
@@ -439,6 +471,9 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
                      conflicts.load());
 
     // --- node buckets per partition, global order preserved (utils.c:1272-1297)
+    // buckets hold PACKED indices; packed order == ascending global id
+    // (sel is ascending), so region content order is rank-invariant — the
+    // property the cross-rank halo slices rely on.
     std::vector<std::vector<int64_t>> pure(P);
     std::vector<std::vector<std::vector<int64_t>>> to_b(P), from_b(P);
     for (int p = 0; p < P; ++p) { to_b[p].resize(P); from_b[p].resize(P); }
@@ -448,8 +483,10 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
         else { to_b[hp][tp].push_back(i); from_b[tp][hp].push_back(i); }
     }
 
-    // markers + global_ids (utils.c:1102-1154)
+    // markers + global_ids (utils.c:1102-1154); global_ids hold PACKED
+    // indices here and are translated through sel at the end
     for (int p = 0; p < P; ++p) {
+        if (focus >= 0 && p != focus) continue;
         Part& pt = g->parts[p];
         pt.markers.reserve(2 * P + 1);
         pt.markers.push_back(0);
@@ -478,6 +515,7 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
 
     DM_T("pt:classify");
     for (int p = 0; p < P; ++p) {
+        if (focus >= 0 && p != focus) continue;
         Part& pt = g->parts[p];
         std::fill(g2l_node.begin(), g2l_node.end(), (int64_t)-1);
         for (int64_t i = 0; i < (int64_t)pt.global_ids.size(); ++i)
@@ -637,6 +675,13 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
                      pt.center_perm, pt.center_row_ptr);
         DM_T("pt:bonds+lines");
     }
+    // translate packed indices -> global atom ids
+    if (sel) {
+        for (int p = 0; p < P; ++p) {
+            Part& pt = g->parts[p];
+            for (auto& gid : pt.global_ids) gid = sel[gid];
+        }
+    }
     return 0;
 }
 
@@ -647,23 +692,97 @@ int build_partitions(dm_graph* g, const double* frac, const double* lat,
 // ---------------------------------------------------------------------------
 extern "C" {
 
+int dm_graph_build_focus(const double* frac_coords, const double* lattice,
+                         const int64_t* pbc, int64_t n_atoms, double cutoff,
+                         double bond_cutoff, double tol,
+                         int32_t num_partitions, int32_t num_threads,
+                         int32_t use_bond_graph, int32_t focus,
+                         dm_graph** out) {
+    if (num_partitions <= 0) { set_err("num_partitions must be >= 1"); return -2; }
+    if (num_threads <= 0) num_threads = 1;
+    if (focus >= num_partitions) { set_err("focus >= num_partitions"); return -2; }
+    auto* g = new dm_graph();
+    g->n_atoms = n_atoms;
+
+    Walls walls;
+    int rc = compute_walls(frac_coords, lattice, n_atoms, num_partitions,
+                           cutoff, bond_cutoff, use_bond_graph != 0, walls);
+    if (rc != 0) { delete g; return rc; }
+
+    // focused slab build: keep only atoms within (cutoff + bond_cutoff +
+    // 1 A) of the focus partition's slab (fractional margin via the
+    // inverse-lattice column norm), with a periodic wrap arm for the
+    // first/last slabs.  Everything relevant to the focus partition —
+    // its atoms, its ghosts (<= cutoff outside) and the src atoms of its
+    // ghost BDEs (<= cutoff + bond_cutoff outside) — lies inside; the
+    // slab-width check guarantees no atom interacts across a full slab.
+    rvec<int64_t> sel;
+    rvec<double> fsub;
+    const double* f_use = frac_coords;
+    int64_t n_use = n_atoms;
+    const int64_t* sel_p = nullptr;
+    if (focus >= 0 && num_partitions > 1) {
+        const int d = walls.dim;
+        // || column d of L^{-1} ||: bound on d(frac_d)/d(cart)
+        double det = lattice[0] * (lattice[4] * lattice[8] - lattice[5] * lattice[7])
+                   - lattice[1] * (lattice[3] * lattice[8] - lattice[5] * lattice[6])
+                   + lattice[2] * (lattice[3] * lattice[7] - lattice[4] * lattice[6]);
+        double ic0, ic1, ic2;
+        if (d == 0) { ic0 = (lattice[4]*lattice[8]-lattice[5]*lattice[7]);
+                      ic1 = (lattice[5]*lattice[6]-lattice[3]*lattice[8]);
+                      ic2 = (lattice[3]*lattice[7]-lattice[4]*lattice[6]); }
+        else if (d == 1) { ic0 = (lattice[2]*lattice[7]-lattice[1]*lattice[8]);
+                           ic1 = (lattice[0]*lattice[8]-lattice[2]*lattice[6]);
+                           ic2 = (lattice[1]*lattice[6]-lattice[0]*lattice[7]); }
+        else { ic0 = (lattice[1]*lattice[5]-lattice[2]*lattice[4]);
+               ic1 = (lattice[2]*lattice[3]-lattice[0]*lattice[5]);
+               ic2 = (lattice[0]*lattice[4]-lattice[1]*lattice[3]); }
+        const double colnorm = std::sqrt(ic0*ic0 + ic1*ic1 + ic2*ic2) / std::abs(det);
+        const double m = (cutoff + bond_cutoff + 1.0) * colnorm;
+        const double lo = (focus == 0 ? walls.fmin : walls.w[focus - 1]) - m;
+        const double hi = (focus == num_partitions - 1 ? walls.fmax
+                                                       : walls.w[focus]) + m;
+        const bool wrap = pbc[d] != 0;
+        sel.reserve(n_atoms / num_partitions * 2);
+        for (int64_t i = 0; i < n_atoms; ++i) {
+            const double f = frac_coords[3 * i + d];
+            bool in = (f >= lo && f <= hi);
+            if (!in && wrap) in = (f - 1.0 >= lo) || (f + 1.0 <= hi);
+            if (in) sel.push_back(i);
+        }
+        const int64_t ns = (int64_t)sel.size();
+        fsub.resize(3 * ns);
+#pragma omp parallel for num_threads(num_threads) schedule(static)
+        for (int64_t i = 0; i < ns; ++i) {
+            fsub[3 * i] = frac_coords[3 * sel[i]];
+            fsub[3 * i + 1] = frac_coords[3 * sel[i] + 1];
+            fsub[3 * i + 2] = frac_coords[3 * sel[i] + 2];
+        }
+        f_use = fsub.data();
+        n_use = ns;
+        sel_p = sel.data();
+    }
+
+    rc = build_neighbor_list(g, f_use, lattice, pbc, n_use, cutoff,
+                             bond_cutoff, tol, num_threads, sel_p);
+    if (rc == 0)
+        rc = build_partitions(g, f_use, sel_p, n_use, walls, cutoff,
+                              bond_cutoff, num_partitions, num_threads,
+                              use_bond_graph != 0,
+                              num_partitions > 1 ? focus : -1);
+    if (rc != 0) { delete g; return rc; }
+    *out = g;
+    return 0;
+}
+
 int dm_graph_build(const double* frac_coords, const double* lattice,
                    const int64_t* pbc, int64_t n_atoms, double cutoff,
                    double bond_cutoff, double tol, int32_t num_partitions,
                    int32_t num_threads, int32_t use_bond_graph,
                    dm_graph** out) {
-    if (num_partitions <= 0) { set_err("num_partitions must be >= 1"); return -2; }
-    if (num_threads <= 0) num_threads = 1;
-    auto* g = new dm_graph();
-    g->n_atoms = n_atoms;
-    int rc = build_neighbor_list(g, frac_coords, lattice, pbc, n_atoms,
-                                 cutoff, bond_cutoff, tol, num_threads);
-    if (rc == 0)
-        rc = build_partitions(g, frac_coords, lattice, cutoff, bond_cutoff,
-                              num_partitions, num_threads, use_bond_graph != 0);
-    if (rc != 0) { delete g; return rc; }
-    *out = g;
-    return 0;
+    return dm_graph_build_focus(frac_coords, lattice, pbc, n_atoms, cutoff,
+                                bond_cutoff, tol, num_partitions, num_threads,
+                                use_bond_graph, -1, out);
 }
 
 int dm_graph_global_view(const dm_graph* g, dm_global_view* out) {

@@ -105,6 +105,7 @@ class Distributed:
         use_bond_graph: bool = False,
         num_threads: int = 1,
         backend=None,
+        focus_partition: int = -1,
     ) -> "Distributed":
         """Partition a periodic structure and build the Distributed graph.
 
@@ -129,7 +130,7 @@ class Distributed:
                 cart_coords, float(cutoff), np.asarray(pbc, dtype=np.int64),
                 lattice_matrix, int(num_partitions), float(three_body_cutoff),
                 float(tol), int(num_threads), bool(use_bond_graph), frac_coords,
-                return_csr=True,
+                return_csr=True, focus=int(focus_partition),
             )
             out, csr_parts = out
         else:

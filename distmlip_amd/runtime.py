@@ -170,6 +170,11 @@ class SpmdEngine:
     # -- graph ------------------------------------------------------------
 
     def build_graph(self, structure) -> Distributed:
+        # focused slab build: each rank builds only its slab + margin —
+        # per-rank graph cost stays O(atoms per slab), the scaling-critical
+        # property (a full build per rank would grow with TOTAL atoms)
+        focus = self.rank if (self.world > 1 and self.graph_backend is None) \
+            else -1
         return Distributed.create_distributed(
             cart_coords=structure.cart_coords,
             frac_coords=structure.frac_coords,
@@ -178,7 +183,8 @@ class SpmdEngine:
             cutoff=self.config.cutoff,
             three_body_cutoff=self.config.three_body_cutoff,
             use_bond_graph=self.use_bond_graph,
-            num_threads=self.threads, backend=self.graph_backend)
+            num_threads=self.threads, backend=self.graph_backend,
+            focus_partition=focus)
 
     # -- one E+F step ------------------------------------------------------
 

@@ -55,6 +55,28 @@ int dm_graph_build(const double* frac_coords,
                    int32_t use_bond_graph,
                    dm_graph** out);
 
+/* Focused build (the SPMD per-rank fast path): identical walls/semantics,
+ * but only atoms within (cutoff + bond_cutoff + 1 A) of partition `focus`'s
+ * slab enter the neighbor list, and only that partition's outputs are
+ * materialized — per-rank cost stays O(atoms per slab) instead of
+ * O(total).  Region content orders are rank-invariant (ascending global
+ * atom id; canonical per-center edge order), so marker halo slices align
+ * across ranks built with different focuses.  With focus = -1 behaves
+ * exactly like dm_graph_build.  NOTE: with focus >= 0 the global-view edge
+ * arrays and per-partition edge_gids are SUBSET-local ids. */
+int dm_graph_build_focus(const double* frac_coords,
+                         const double* lattice,
+                         const int64_t* pbc,
+                         int64_t n_atoms,
+                         double cutoff,
+                         double bond_cutoff,
+                         double tol,
+                         int32_t num_partitions,
+                         int32_t num_threads,
+                         int32_t use_bond_graph,
+                         int32_t focus,
+                         dm_graph** out);
+
 typedef struct {
     int64_t n_atoms;
     int64_t n_edges;            /* global directed edges */
