@@ -63,11 +63,24 @@ def halo_plan(markers, rank: int, P: int) -> List[Tuple[int, int, int, int, int]
     return plan
 
 
+import os as _os
+
+_GLOO_HALO_GROUP = None   # emergency fallback: DM_HALO_GLOO=1 stages halo
+                          # slices through host memory over a gloo group
+
+
 def _exchange(feat: torch.Tensor, plan, reverse: bool = False) -> dict:
     """Grouped p2p exchange (batch_isend_irecv: one RCCL group, deadlock-
     free by construction); returns {peer: recv_buffer}.  reverse=True swaps
     roles (send my recv-slices, receive for my send-slices) — the backward
     direction."""
+    if _GLOO_HALO_GROUP is not None and feat.device.type == "cuda":
+        recvs_cpu = _exchange_via(feat.cpu(), plan, reverse, _GLOO_HALO_GROUP)
+        return {q: b.to(feat.device) for q, b in recvs_cpu.items()}
+    return _exchange_via(feat, plan, reverse, None)
+
+
+def _exchange_via(feat, plan, reverse, group):
     recvs, p2p, keep = {}, [], []
     for (q, ss, se, rs, re) in plan:
         c, d = (ss, se) if reverse else (rs, re)
@@ -75,13 +88,13 @@ def _exchange(feat: torch.Tensor, plan, reverse: bool = False) -> dict:
             buf = torch.empty((d - c,) + tuple(feat.shape[1:]),
                               dtype=feat.dtype, device=feat.device)
             recvs[q] = buf
-            p2p.append(dist.P2POp(dist.irecv, buf, q))
+            p2p.append(dist.P2POp(dist.irecv, buf, q, group=group))
     for (q, ss, se, rs, re) in plan:
         a, b = (rs, re) if reverse else (ss, se)
         if b > a:
             sbuf = feat[a:b].contiguous()
             keep.append(sbuf)                        # alive until wait
-            p2p.append(dist.P2POp(dist.isend, sbuf, q))
+            p2p.append(dist.P2POp(dist.isend, sbuf, q, group=group))
     if p2p:
         for r in dist.batch_isend_irecv(p2p):
             r.wait()
@@ -145,6 +158,10 @@ class SpmdEngine:
         self.rank = dist.get_rank() if dist.is_initialized() else 0
         assert world == 1 or dist.is_initialized(), \
             "world > 1 needs an initialized torch.distributed process group"
+        global _GLOO_HALO_GROUP
+        if world > 1 and _os.environ.get("DM_HALO_GLOO") == "1" \
+                and _GLOO_HALO_GROUP is None:
+            _GLOO_HALO_GROUP = dist.new_group(backend="gloo")
         self.world = world
         self.config = core.config
         self.use_bond_graph = use_bond_graph and core.config.use_bond_graph
