@@ -157,9 +157,7 @@ def main():
     # host regresses 5x (NUMA + team spawn) — cap unless told otherwise
     threads = args.threads or max(1, min(32, (os.cpu_count() or 8) // max(world, 1)))
 
-    from distmlip_amd.chgnet import CHGNet_Dist
     from distmlip_amd.model import CHGNetCore
-    from distmlip_amd.pes import Potential_Dist
     from distmlip_amd.structures import workload
 
     s = workload(args.workload, n_gpus=n_gpus if args.workload == "li100k" else 1)

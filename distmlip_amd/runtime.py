@@ -33,7 +33,6 @@ import torch
 import torch.distributed as dist
 
 from distmlip_amd.chgnet import (
-    CHGNet_Dist,
     PartitionData,
     gated_mlp_split3,
     gated_mlp_split4,
@@ -41,7 +40,6 @@ from distmlip_amd.chgnet import (
 from distmlip_amd.dist import Distributed
 from distmlip_amd.model import (
     CHGNetCore,
-    bond_expansion_from_dist,
     compute_theta,
     fourier_expansion,
 )
