@@ -184,7 +184,7 @@ class SpmdEngine:
 
     # -- graph ------------------------------------------------------------
 
-    def build_graph(self, structure) -> Distributed:
+    def build_graph(self, structure, skin: float = 0.0) -> Distributed:
         # focused slab build: each rank builds only its slab + margin —
         # per-rank graph cost stays O(atoms per slab), the scaling-critical
         # property (a full build per rank would grow with TOTAL atoms)
@@ -195,16 +195,55 @@ class SpmdEngine:
             frac_coords=structure.frac_coords,
             lattice_matrix=structure.lattice,
             num_partitions=self.world, pbc=structure.pbc,
-            cutoff=self.config.cutoff,
-            three_body_cutoff=self.config.three_body_cutoff,
+            cutoff=self.config.cutoff + skin,
+            three_body_cutoff=self.config.three_body_cutoff +
+            (skin if self.use_bond_graph else 0.0),
             use_bond_graph=self.use_bond_graph,
             num_threads=self.threads, backend=self.graph_backend,
             focus_partition=focus)
 
+    # -- Verlet-skin graph reuse across MD steps ----------
+    #
+    # Build once with cutoffs enlarged by `skin`; while no atom has moved
+    # more than skin/2 since the build, the enlarged edge list is a
+    # superset of the true one, and per-step distance MASKS on the shared
+    # message weights make the result EXACTLY what a fresh build would
+    # give: masked terms are exact fp zeros, so every reduction is
+    # bit-identical to the fresh-graph sum.  This removes the per-step
+    # CPU rebuild the reference pays (pes.py:75-85; 31 s at 1M atoms).
+
+    def step_verlet(self, structure, skin: float = 1.0, **kw):
+        rebuild = True
+        if getattr(self, "_vcache", None) is not None:
+            c = self._vcache
+            if c["skin"] == skin and len(c["frac"]) == structure.num_atoms:
+                d = structure.frac_coords - c["frac"]
+                d = (d + 0.5) % 1.0 - 0.5                 # minimal image
+                disp2 = ((d @ structure.lattice) ** 2).sum(1)
+                rebuild = float(disp2.max()) > (skin / 2) ** 2
+        if rebuild:
+            self._vcache = {
+                "skin": skin,
+                "frac": structure.frac_coords.copy(),
+                "dist_info": self.build_graph(structure, skin=skin),
+                "rebuilds": getattr(self, "_vcache", None) and
+                self._vcache["rebuilds"] + 1 or 1,
+            }
+            frac_eff = None
+        else:
+            # atoms that re-wrapped across a periodic boundary since the
+            # build would invalidate the stored edge images; evaluate at
+            # build-frame coordinates + minimal-image displacement instead
+            d = structure.frac_coords - self._vcache["frac"]
+            frac_eff = self._vcache["frac"] + ((d + 0.5) % 1.0 - 0.5)
+        return self.step(structure, dist_info=self._vcache["dist_info"],
+                         mask_cutoffs=True, frac_override=frac_eff, **kw)
+
     # -- one E+F step ------------------------------------------------------
 
     def step(self, structure, dist_info: Optional[Distributed] = None,
-             calc_stresses: bool = False):
+             calc_stresses: bool = False, mask_cutoffs: bool = False,
+             frac_override: Optional[np.ndarray] = None):
         """Returns (total_energy_scalar, owned_forces[n_owned,3]).
 
         total_energy includes scale/shift + element refs (pes.py:109-113);
@@ -231,7 +270,9 @@ class SpmdEngine:
             strain.requires_grad_(True)
         lattice = lat0 @ (torch.eye(3, device=dev, dtype=ft) + strain)
 
-        frac_local = torch.tensor(np.asarray(structure.frac_coords)[gids],
+        frac_src = frac_override if frac_override is not None \
+            else structure.frac_coords
+        frac_local = torch.tensor(np.asarray(frac_src)[gids],
                                   dtype=ft, device=dev)
         pos = frac_local @ lattice
         if not pos.requires_grad:
@@ -256,6 +297,13 @@ class SpmdEngine:
             pos, offshift, core.rbf_freq_atom, cfg.cutoff,
             cfg.cutoff_exponent, pd)
 
+        # Verlet-skin mode: the edge list is a superset built at
+        # cutoff+skin; mask the shared message weights by the TRUE cutoffs
+        # so every out-of-range contribution is an exact zero
+        edge_mask = None
+        if mask_cutoffs:
+            edge_mask = (bond_dist.detach() < cfg.cutoff).to(ft).unsqueeze(1)
+
         v = core.atom_embedding(species_local)
         e = core.bond_embedding(bond_expansion)
 
@@ -272,6 +320,10 @@ class SpmdEngine:
             nd_dist = HaloExchange.apply(nd_dist.unsqueeze(1), line_plan).squeeze(1)
             nd_vec = HaloExchange.apply(nd_vec, line_plan)
 
+            bond_mask = None
+            if mask_cutoffs:
+                bond_mask = (nd_dist.detach() <
+                             cfg.three_body_cutoff).to(ft).unsqueeze(1)
             exp3 = ops.rbf_env(nd_dist, core.rbf_freq_bond,
                                cfg.three_body_cutoff, cfg.cutoff_exponent)
             theta = compute_theta(
@@ -296,18 +348,24 @@ class SpmdEngine:
         # the [E,64]/[B,64] weight tensors are recomputed inside each
         # segment from the small [*,9] expansions instead of being held
         # for the whole forward (2x [E,64] = 22 GB at 1M atoms).
+        def _m(w, m):
+            return w if m is None else w * m
+
         if ckpt:
             w_ab = w_bb = None
             w_3b = None
         else:
-            w_ab = core.atom_bond_weights(bond_expansion)
-            w_bb = core.bond_bond_weights(bond_expansion)
-            w_3b = core.threebody_bond_weights(exp3) if use_bg else None
+            w_ab = _m(core.atom_bond_weights(bond_expansion), edge_mask)
+            w_bb = _m(core.bond_bond_weights(bond_expansion), edge_mask)
+            w_3b = _m(core.threebody_bond_weights(exp3), bond_mask) \
+                if use_bg else None
 
         def atom_conv_body(layer_idx_t, v, e, bexp):
             blk = core.atom_convs[int(layer_idx_t)]
-            wbb = w_bb if w_bb is not None else core.bond_bond_weights(bexp)
-            wab = w_ab if w_ab is not None else core.atom_bond_weights(bexp)
+            wbb = w_bb if w_bb is not None else \
+                _m(core.bond_bond_weights(bexp), edge_mask)
+            wab = w_ab if w_ab is not None else \
+                _m(core.atom_bond_weights(bexp), edge_mask)
             e = gated_mlp_split3(blk.edge_mlp, v, e, pd, ops, d,
                                  w=wbb, base=e)
             msg = gated_mlp_split3(blk.node_mlp, v, e, pd, ops, d, w=wab)
@@ -330,10 +388,16 @@ class SpmdEngine:
 
                 def bond_body(n, a, v, e3, _blk=blk):
                     w3 = w_3b if w_3b is not None else \
-                        core.threebody_bond_weights(e3)
-                    msg = gated_mlp_split4(
-                        _blk.bond_mlp, n, a, v, pd, ops, d,
-                        w=ops.gather(w3, pd.l_src, csr=pd.line_src_csr))
+                        _m(core.threebody_bond_weights(e3), bond_mask)
+                    wl = ops.gather(w3, pd.l_src, csr=pd.line_src_csr)
+                    if bond_mask is not None:
+                        # Verlet mode: a message into a MASKED (superset)
+                        # bond would later leak back into a live atom edge
+                        # through bond_to_edge — kill by destination too
+                        wl = wl * ops.gather(bond_mask, pd.l_dst,
+                                             csr=pd.line_dst_csr)
+                    msg = gated_mlp_split4(_blk.bond_mlp, n, a, v, pd, ops, d,
+                                           w=wl)
                     return ops.scatter_lines(msg, pd, base=n)
 
                 n = _ck(bond_body, n, a, v, exp3)
