@@ -69,8 +69,7 @@ def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
     returns base + silu(core2(silu(z_c))) * sigmoid(gate2(silu(z_g))) * w."""
     wcg, bcg = _cg_weight(mlp)
     ws, wd, we = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:]
-    z = ops.gather_add3(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
-    h = F.silu(z)
+    h = ops.gather_add3_act(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
     c = mlp.core2(h[:, :d])
     g = mlp.gate2(h[:, d:])
     return ops.gated_combine(c, g, w, base)
@@ -82,9 +81,8 @@ def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
     wcg, bcg = _cg_weight(mlp)
     w1, w2, wa, wv = (wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:3 * d],
                       wcg[:, 3 * d:])
-    z = ops.gather_add4(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
-                        v @ wv.t(), pd)
-    h = F.silu(z)
+    h = ops.gather_add4_act(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
+                            v @ wv.t(), pd)
     c = mlp.core2(h[:, :d])
     g = mlp.gate2(h[:, d:])
     return ops.gated_combine(c, g, w, base)

@@ -70,12 +70,17 @@ __global__ void k_gather_rows_s(const float* __restrict__ x,
     }
 }
 
+__device__ __forceinline__ float siluf(float x) {
+    return x / (1.0f + expf(-x));
+}
+
 __global__ void k_gather_add3_v4(const float4* __restrict__ zs,
                                  const float4* __restrict__ zd,
                                  const float4* __restrict__ ze,
                                  const int32_t* __restrict__ src,
                                  const int32_t* __restrict__ dst,
                                  float4* __restrict__ out,
+                                 float4* __restrict__ out_act,
                                  int64_t total, int32_t D4) {
     for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          t < total; t += (int64_t)gridDim.x * blockDim.x) {
@@ -84,8 +89,27 @@ __global__ void k_gather_add3_v4(const float4* __restrict__ zs,
         const float4 a = zs[(int64_t)src[row] * D4 + c];
         const float4 b = zd[(int64_t)dst[row] * D4 + c];
         const float4 e = ze[t];
-        out[t] = make_float4(a.x + b.x + e.x, a.y + b.y + e.y,
-                             a.z + b.z + e.z, a.w + b.w + e.w);
+        const float4 z = make_float4(a.x + b.x + e.x, a.y + b.y + e.y,
+                                     a.z + b.z + e.z, a.w + b.w + e.w);
+        out[t] = z;
+        if (out_act)
+            out_act[t] = make_float4(siluf(z.x), siluf(z.y), siluf(z.z),
+                                     siluf(z.w));
+    }
+}
+
+// dz = (go_z ? go_z : 0) + go_h * silu'(z) — fused SiLU backward
+__global__ void k_silu_bwd(const float* __restrict__ go_h,
+                           const float* __restrict__ go_z,
+                           const float* __restrict__ z,
+                           float* __restrict__ dz, int64_t total) {
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const float zv = z[t];
+        const float s = 1.0f / (1.0f + expf(-zv));
+        float g = go_h[t] * (s * (1.0f + zv * (1.0f - s)));
+        if (go_z) g += go_z[t];
+        dz[t] = g;
     }
 }
 
@@ -97,6 +121,7 @@ __global__ void k_gather_add4_v4(const float4* __restrict__ z1,
                                  const int32_t* __restrict__ ldst,
                                  const int32_t* __restrict__ center,
                                  float4* __restrict__ out,
+                                 float4* __restrict__ out_act,
                                  int64_t total, int32_t D4) {
     for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          t < total; t += (int64_t)gridDim.x * blockDim.x) {
@@ -106,8 +131,12 @@ __global__ void k_gather_add4_v4(const float4* __restrict__ z1,
         const float4 b = z2[(int64_t)ldst[row] * D4 + c];
         const float4 e = za[t];
         const float4 v = zv[(int64_t)center[row] * D4 + c];
-        out[t] = make_float4(a.x + b.x + e.x + v.x, a.y + b.y + e.y + v.y,
-                             a.z + b.z + e.z + v.z, a.w + b.w + e.w + v.w);
+        const float4 z = make_float4(a.x + b.x + e.x + v.x, a.y + b.y + e.y + v.y,
+                                     a.z + b.z + e.z + v.z, a.w + b.w + e.w + v.w);
+        out[t] = z;
+        if (out_act)
+            out_act[t] = make_float4(siluf(z.x), siluf(z.y), siluf(z.z),
+                                     siluf(z.w));
     }
 }
 
@@ -417,13 +446,14 @@ int dm_gather_rows_f32(const float* x, const int32_t* idx, float* out,
 
 int dm_gather_add3_f32(const float* zs, const float* zd, const float* ze,
                        const int32_t* src, const int32_t* dst, float* out,
-                       int64_t E, int64_t D, uint64_t stream) {
+                       float* out_act, int64_t E, int64_t D,
+                       uint64_t stream) {
     hipStream_t s = (hipStream_t)stream;
     if (D % 4 != 0) { g_err = "gather_add3 requires D % 4 == 0"; return -1; }
     const int64_t total = E * (D / 4);
     k_gather_add3_v4<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
         (const float4*)zs, (const float4*)zd, (const float4*)ze, src, dst,
-        (float4*)out, total, (int32_t)(D / 4));
+        (float4*)out, (float4*)out_act, total, (int32_t)(D / 4));
     DM_CHECK_LAUNCH();
     return 0;
 }
@@ -431,14 +461,15 @@ int dm_gather_add3_f32(const float* zs, const float* zd, const float* ze,
 int dm_gather_add4_f32(const float* z1, const float* z2, const float* za,
                        const float* zv, const int32_t* lsrc,
                        const int32_t* ldst, const int32_t* center, float* out,
-                       int64_t L, int64_t D, uint64_t stream) {
+                       float* out_act, int64_t L, int64_t D,
+                       uint64_t stream) {
     hipStream_t s = (hipStream_t)stream;
     if (D % 4 != 0) { g_err = "gather_add4 requires D % 4 == 0"; return -1; }
     const int64_t total = L * (D / 4);
     k_gather_add4_v4<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(
         (const float4*)z1, (const float4*)z2, (const float4*)za,
-        (const float4*)zv, lsrc, ldst, center, (float4*)out, total,
-        (int32_t)(D / 4));
+        (const float4*)zv, lsrc, ldst, center, (float4*)out,
+        (float4*)out_act, total, (int32_t)(D / 4));
     DM_CHECK_LAUNCH();
     return 0;
 }
@@ -539,6 +570,15 @@ int dm_rbf_env_bwd_f32(const float* go_exp, const float* d, const float* freqs,
     hipStream_t s = (hipStream_t)stream;
     k_rbf_env_bwd<<<nblocks(M, BLOCK), BLOCK, 0, s>>>(
         go_exp, d, freqs, cutoff, pexp, nrbf, gd, M);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_silu_bwd_f32(const float* go_h, const float* go_z, const float* z,
+                    float* dz, int64_t total, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    k_silu_bwd<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(go_h, go_z, z, dz,
+                                                       total);
     DM_CHECK_LAUNCH();
     return 0;
 }

@@ -31,11 +31,13 @@ def hip_lib() -> ctypes.CDLL:
         lib.dm_gather_rows_f32.restype = c_int32
         lib.dm_gather_rows_f32.argtypes = [fp, ip, fp, c_int64, c_int64, c_uint64]
         lib.dm_gather_add3_f32.restype = c_int32
-        lib.dm_gather_add3_f32.argtypes = [fp, fp, fp, ip, ip, fp, c_int64,
-                                           c_int64, c_uint64]
+        lib.dm_gather_add3_f32.argtypes = [fp, fp, fp, ip, ip, fp, fp,
+                                           c_int64, c_int64, c_uint64]
         lib.dm_gather_add4_f32.restype = c_int32
         lib.dm_gather_add4_f32.argtypes = [fp, fp, fp, fp, ip, ip, ip, fp,
-                                           c_int64, c_int64, c_uint64]
+                                           fp, c_int64, c_int64, c_uint64]
+        lib.dm_silu_bwd_f32.restype = c_int32
+        lib.dm_silu_bwd_f32.argtypes = [fp, fp, fp, fp, c_int64, c_uint64]
         lib.dm_seg_sum_f32.restype = c_int32
         lib.dm_seg_sum_f32.argtypes = [fp, ip, fp, fp, c_int64, c_int64,
                                        c_uint64]
@@ -144,51 +146,79 @@ class _Gather(torch.autograd.Function):
         return gx, None, None, None
 
 
+def raw_silu_bwd(go_h, go_z, z):
+    """dz = (go_z or 0) + go_h * silu'(z), one fused pass."""
+    dz = torch.empty_like(z)
+    _check(hip_lib().dm_silu_bwd_f32(
+        _fp(go_h), _fp(go_z) if go_z is not None else None, _fp(z), _fp(dz),
+        z.numel(), _stream()), "dm_silu_bwd_f32")
+    return dz
+
+
 class _GatherAdd3(torch.autograd.Function):
+    """Emits (z, silu(z)) in one kernel; z doubles as the saved activation
+    for the fused silu backward."""
+
     @staticmethod
     def forward(ctx, zs, zd, ze, src, dst, src_perm, src_row_ptr, row_ptr):
         _chk_f32(zs, zd, ze)
-        ctx.save_for_backward(src_perm, src_row_ptr, row_ptr)
         ctx.n_nodes = zs.shape[0]
-        out = torch.empty_like(ze)
+        z = torch.empty_like(ze)
+        h = torch.empty_like(ze)
         _check(hip_lib().dm_gather_add3_f32(
-            _fp(zs), _fp(zd), _fp(ze), _ip(src), _ip(dst), _fp(out),
+            _fp(zs), _fp(zd), _fp(ze), _ip(src), _ip(dst), _fp(z), _fp(h),
             ze.shape[0], ze.shape[1], _stream()), "dm_gather_add3_f32")
-        return out
+        ctx.save_for_backward(src_perm, src_row_ptr, row_ptr, z)
+        return z, h
 
     @staticmethod
-    def backward(ctx, grad):
-        src_perm, src_row_ptr, row_ptr = ctx.saved_tensors
-        grad = grad.contiguous()
-        gzs = raw_seg_sum_gather(grad, src_perm, src_row_ptr, ctx.n_nodes)
-        gzd = raw_seg_sum(grad, row_ptr, ctx.n_nodes)
-        return gzs, gzd, grad, None, None, None, None, None
+    def backward(ctx, go_z, go_h):
+        src_perm, src_row_ptr, row_ptr, z = ctx.saved_tensors
+        if go_h is not None:
+            dz = raw_silu_bwd(go_h.contiguous(),
+                              go_z.contiguous() if go_z is not None else None,
+                              z)
+        else:
+            dz = go_z.contiguous()
+        gzs = raw_seg_sum_gather(dz, src_perm, src_row_ptr, ctx.n_nodes)
+        gzd = raw_seg_sum(dz, row_ptr, ctx.n_nodes)
+        return gzs, gzd, dz, None, None, None, None, None
 
 
 class _GatherAdd4(torch.autograd.Function):
+    """Emits (z, silu(z)) over the line-graph 3-gather-add."""
+
     @staticmethod
     def forward(ctx, z1, z2, za, zv, pd):
         _chk_f32(z1, z2, za, zv)
         ctx.pd = pd
         ctx.n_bonds = z1.shape[0]
         ctx.n_nodes = zv.shape[0]
-        out = torch.empty_like(za)
+        z = torch.empty_like(za)
+        h = torch.empty_like(za)
         _check(hip_lib().dm_gather_add4_f32(
             _fp(z1), _fp(z2), _fp(za), _fp(zv), _ip(pd.l_src), _ip(pd.l_dst),
-            _ip(pd.center), _fp(out), za.shape[0], za.shape[1], _stream()),
-            "dm_gather_add4_f32")
-        return out
+            _ip(pd.center), _fp(z), _fp(h), za.shape[0], za.shape[1],
+            _stream()), "dm_gather_add4_f32")
+        ctx.save_for_backward(z)
+        return z, h
 
     @staticmethod
-    def backward(ctx, grad):
+    def backward(ctx, go_z, go_h):
         pd = ctx.pd
-        grad = grad.contiguous()
-        gz1 = raw_seg_sum_gather(grad, pd.line_src_perm, pd.line_src_row_ptr,
+        (z,) = ctx.saved_tensors
+        if go_h is not None:
+            dz = raw_silu_bwd(go_h.contiguous(),
+                              go_z.contiguous() if go_z is not None else None,
+                              z)
+        else:
+            dz = go_z.contiguous()
+        gz1 = raw_seg_sum_gather(dz, pd.line_src_perm, pd.line_src_row_ptr,
                                  ctx.n_bonds)
-        gz2 = raw_seg_sum(grad, pd.line_row_ptr, ctx.n_bonds)
-        gzv = raw_seg_sum_gather(grad, pd.center_perm, pd.center_row_ptr,
+        gz2 = raw_seg_sum(dz, pd.line_row_ptr, ctx.n_bonds)
+        gzv = raw_seg_sum_gather(dz, pd.center_perm, pd.center_row_ptr,
                                  ctx.n_nodes)
-        return gz1, gz2, grad, gzv, None
+        return gz1, gz2, dz, gzv, None
 
 
 class _SegSum(torch.autograd.Function):
@@ -318,14 +348,17 @@ class HipOps:
                                "for its deterministic backward")
         return _Gather.apply(x.contiguous(), idx, csr[0], csr[1])
 
-    def gather_add3(self, zs, zd, ze, pd):
-        return _GatherAdd3.apply(zs.contiguous(), zd.contiguous(),
-                                 ze.contiguous(), pd.src, pd.dst,
-                                 pd.src_perm, pd.src_row_ptr, pd.row_ptr)
+    def gather_add3_act(self, zs, zd, ze, pd):
+        """silu(zs[src] + zd[dst] + ze), silu fused into the gather kernel."""
+        _z, h = _GatherAdd3.apply(zs.contiguous(), zd.contiguous(),
+                                  ze.contiguous(), pd.src, pd.dst,
+                                  pd.src_perm, pd.src_row_ptr, pd.row_ptr)
+        return h
 
-    def gather_add4(self, z1, z2, za, zv, pd):
-        return _GatherAdd4.apply(z1.contiguous(), z2.contiguous(),
-                                 za.contiguous(), zv.contiguous(), pd)
+    def gather_add4_act(self, z1, z2, za, zv, pd):
+        _z, h = _GatherAdd4.apply(z1.contiguous(), z2.contiguous(),
+                                  za.contiguous(), zv.contiguous(), pd)
+        return h
 
     def scatter_edges(self, msg, pd, base=None):
         return _SegSum.apply(msg, pd.dst, pd.row_ptr, pd.n_atoms, base)

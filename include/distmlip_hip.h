@@ -34,17 +34,24 @@ int dm_gather_rows_f32(const float* x, const int32_t* idx, float* out,
                        int64_t n_out, int64_t D, uint64_t stream);
 
 /* out[e, :] = zs[src[e], :] + zd[dst[e], :] + ze[e, :]  (fused 2-gather-add:
- * the split-linear form of GatedMLP(cat(v_src, v_dst, e))) */
+ * the split-linear form of GatedMLP(cat(v_src, v_dst, e))).  If out_act is
+ * non-null also emits silu(out) in the same pass (the gated-MLP hidden
+ * activation). */
 int dm_gather_add3_f32(const float* zs, const float* zd, const float* ze,
                        const int32_t* src, const int32_t* dst, float* out,
-                       int64_t E, int64_t D, uint64_t stream);
+                       float* out_act, int64_t E, int64_t D, uint64_t stream);
 
 /* out[l, :] = z1[lsrc[l], :] + z2[ldst[l], :] + za[l, :] + zv[center[l], :]
- * (bond/line-graph GatedMLP(cat(n_b1, n_b2, a, v_center))) */
+ * (bond/line-graph GatedMLP(cat(n_b1, n_b2, a, v_center))); optional fused
+ * silu as above. */
 int dm_gather_add4_f32(const float* z1, const float* z2, const float* za,
                        const float* zv, const int32_t* lsrc,
                        const int32_t* ldst, const int32_t* center, float* out,
-                       int64_t L, int64_t D, uint64_t stream);
+                       float* out_act, int64_t L, int64_t D, uint64_t stream);
+
+/* dz = (go_z ? go_z : 0) + go_h * silu'(z) — backward of the fused silu */
+int dm_silu_bwd_f32(const float* go_h, const float* go_z, const float* z,
+                    float* dz, int64_t total, uint64_t stream);
 
 /* out[n, :] = (base ? base[n, :] : 0) + sum_{j in [row_ptr[n], row_ptr[n+1])}
  *             msg[j, :]

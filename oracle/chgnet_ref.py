@@ -193,11 +193,12 @@ class CpuRefOps:
     def gather(self, x, idx, csr=None):
         return x[idx]
 
-    def gather_add3(self, zs, zd, ze, pd):
-        return zs[pd.src] + zd[pd.dst] + ze
+    def gather_add3_act(self, zs, zd, ze, pd):
+        return torch.nn.functional.silu(zs[pd.src] + zd[pd.dst] + ze)
 
-    def gather_add4(self, z1, z2, za, zv, pd):
-        return z1[pd.l_src] + z2[pd.l_dst] + za + zv[pd.center]
+    def gather_add4_act(self, z1, z2, za, zv, pd):
+        return torch.nn.functional.silu(
+            z1[pd.l_src] + z2[pd.l_dst] + za + zv[pd.center])
 
     def _scatter(self, msg, idx, n_out, base):
         out = torch.zeros((n_out,) + tuple(msg.shape[1:]), dtype=msg.dtype,
