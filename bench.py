@@ -46,6 +46,10 @@ def parse_args():
                     choices=["li100k", "si1m", "si1k"])
     ap.add_argument("--threads", type=int, default=0,
                     help="graph-builder threads (0 = cpu_count/world)")
+    ap.add_argument("--verlet", type=float, default=0.0, metavar="SKIN",
+                    help="MD mode: random-walk the structure each step and "
+                         "reuse the graph via Verlet-skin SKIN (A); the "
+                         "headline line always uses per-step rebuilds")
     ap.add_argument("--no-bond-graph", action="store_true")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--breakdown", action="store_true",
@@ -176,6 +180,18 @@ def main():
     def step():
         return engine.step(s)
 
+    if args.verlet > 0:
+        walk_rng = np.random.default_rng(1234)  # SAME walk on every rank
+        inv_lat = np.linalg.inv(s.lattice)
+
+        def step():  # noqa: F811
+            # MD-style random walk (~0.05 A/step) + Verlet-skin graph reuse
+            cart = s.frac_coords @ s.lattice
+            cart += walk_rng.normal(0, 0.03, size=cart.shape)
+            f = cart @ inv_lat
+            s.frac_coords = np.mod(np.mod(f, 1.0), 1.0)
+            return engine.step_verlet(s, skin=args.verlet)
+
     if world == 1 and args.breakdown:
         stages = {}
 
@@ -251,6 +267,9 @@ def main():
             "data": "synthetic",
             "config": {
                 "workload": args.workload,
+                "verlet_skin": args.verlet or None,
+                "verlet_rebuilds": (getattr(engine, "_vcache", None) or
+                                    {}).get("rebuilds"),
                 "n_atoms": int(total_atoms),
                 "cutoff": 6.0,
                 "three_body_cutoff": 3.0,
