@@ -176,6 +176,46 @@ __global__ void k_seg_sum_v4(const float4* __restrict__ msg,
     }
 }
 
+// unroll-4 variant of the D4<=16 segmented sum: keeps 4 independent row
+// reads in flight per group iteration (A/B candidate vs k_seg_sum_v4)
+__global__ void k_seg_sum_v4_u4(const float4* __restrict__ msg,
+                                const int32_t* __restrict__ rp,
+                                const float4* __restrict__ base,
+                                float4* __restrict__ out,
+                                int64_t N, int32_t D4) {
+    const int lane16 = threadIdx.x & 15;
+    const int group = threadIdx.x >> 4;
+    const int groups_per_block = blockDim.x >> 4;
+    for (int64_t row = blockIdx.x * (int64_t)groups_per_block + group;
+         row < N; row += (int64_t)gridDim.x * groups_per_block) {
+        if (lane16 < D4) {
+            double ax = 0, ay = 0, az = 0, aw = 0;
+            if (base) {
+                const float4 b = base[row * D4 + lane16];
+                ax = b.x; ay = b.y; az = b.z; aw = b.w;
+            }
+            const int32_t lo = rp[row], hi = rp[row + 1];
+            int32_t j = lo;
+            for (; j + 4 <= hi; j += 4) {
+                const float4 m0 = msg[(int64_t)(j + 0) * D4 + lane16];
+                const float4 m1 = msg[(int64_t)(j + 1) * D4 + lane16];
+                const float4 m2 = msg[(int64_t)(j + 2) * D4 + lane16];
+                const float4 m3 = msg[(int64_t)(j + 3) * D4 + lane16];
+                ax += m0.x; ay += m0.y; az += m0.z; aw += m0.w;
+                ax += m1.x; ay += m1.y; az += m1.z; aw += m1.w;
+                ax += m2.x; ay += m2.y; az += m2.z; aw += m2.w;
+                ax += m3.x; ay += m3.y; az += m3.z; aw += m3.w;
+            }
+            for (; j < hi; ++j) {
+                const float4 m = msg[(int64_t)j * D4 + lane16];
+                ax += m.x; ay += m.y; az += m.z; aw += m.w;
+            }
+            out[row * D4 + lane16] = make_float4((float)ax, (float)ay,
+                                                 (float)az, (float)aw);
+        }
+    }
+}
+
 // generic-D scalar path: one thread per (row, col); cols of one row sit on
 // consecutive threads so each j-iteration is a coalesced D*4-byte read
 __global__ void k_seg_sum_s(const float* __restrict__ msg,
@@ -474,14 +514,24 @@ int dm_gather_add4_f32(const float* z1, const float* z2, const float* za,
     return 0;
 }
 
+static int g_seg_variant = []() {
+    const char* v = getenv("DM_SEG_VARIANT");
+    return v ? atoi(v) : 1;            // 1 = unroll-4 (default), 0 = plain
+}();
+
 int dm_seg_sum_f32(const float* msg, const int32_t* row_ptr,
                    const float* base, float* out, int64_t N, int64_t D,
                    uint64_t stream) {
     hipStream_t s = (hipStream_t)stream;
     if (D % 4 == 0 && D / 4 <= 16) {
-        k_seg_sum_v4<<<nblocks(N, 16), BLOCK, 0, s>>>(
-            (const float4*)msg, row_ptr, (const float4*)base, (float4*)out,
-            N, (int32_t)(D / 4));
+        if (g_seg_variant == 1)
+            k_seg_sum_v4_u4<<<nblocks(N, 16), BLOCK, 0, s>>>(
+                (const float4*)msg, row_ptr, (const float4*)base,
+                (float4*)out, N, (int32_t)(D / 4));
+        else
+            k_seg_sum_v4<<<nblocks(N, 16), BLOCK, 0, s>>>(
+                (const float4*)msg, row_ptr, (const float4*)base,
+                (float4*)out, N, (int32_t)(D / 4));
     } else {
         k_seg_sum_s<<<nblocks(N * D, BLOCK), BLOCK, 0, s>>>(
             msg, row_ptr, base, out, N, (int32_t)D);
