@@ -1,0 +1,113 @@
+"""Size-independent physics invariants at BASELINE full scale (tier ③:
+properties replace the oracle where the oracle cannot run at size).
+All on the product GPU path (SpmdEngine world=1, HIP kernels)."""
+import copy
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs HIP GPU")
+
+
+def _engine():
+    from distmlip_amd.model import CHGNetCore
+    from distmlip_amd.runtime import SpmdEngine
+    core = CHGNetCore.seeded(seed=0).float()
+    return SpmdEngine(core, world=1, threads=8)
+
+
+def _full_forces(s, out):
+    F = np.zeros((s.num_atoms, 3))
+    F[out["global_ids_owned"]] = out["forces_owned"].cpu().numpy()
+    return F
+
+
+@requires_gpu
+def test_li100k_invariants():
+    """config #2 scale (101,306 atoms): force sum ~ 0, translation
+    invariance of E and F, permutation invariance."""
+    from distmlip_amd.structures import workload
+    s = workload("li100k")
+    eng = _engine()
+    out = eng.step(s)
+    E = out["energy"].item()
+    F = _full_forces(s, out)
+
+    # Newton's third law: total force vanishes
+    assert np.abs(F.sum(0)).max() < 2e-2, F.sum(0)
+
+    # rigid translation: same energy, same forces
+    s2 = copy.deepcopy(s)
+    s2.frac_coords = np.mod(s2.frac_coords + np.array([0.213, 0.377, 0.119]),
+                            1.0)
+    out2 = eng.step(s2)
+    E2 = out2["energy"].item()
+    F2 = _full_forces(s2, out2)
+    assert abs(E2 - E) < 5e-3 * max(1.0, abs(E)), (E, E2)
+    assert np.abs(F2 - F).max() < 5e-3, np.abs(F2 - F).max()
+
+    # permutation invariance: shuffled atom order, same physics
+    rng = np.random.default_rng(7)
+    perm = rng.permutation(s.num_atoms)
+    s3 = copy.deepcopy(s)
+    s3.frac_coords = s.frac_coords[perm]
+    s3.species = s.species[perm]
+    out3 = eng.step(s3)
+    F3 = _full_forces(s3, out3)
+    assert abs(out3["energy"].item() - E) < 5e-3 * max(1.0, abs(E))
+    assert np.abs(F3 - F[perm]).max() < 5e-3
+
+
+@requires_gpu
+def test_li100k_partition_invariance_on_one_gpu():
+    """P=2 single-process multi-partition on ONE GPU at ~100k atoms must
+    match the world=1 engine (full-size partition invariance on the
+    product path)."""
+    from distmlip_amd.chgnet import CHGNet_Dist
+    from distmlip_amd.model import CHGNetCore
+    from distmlip_amd.pes import Potential_Dist
+    from distmlip_amd.structures import workload
+    s = workload("li100k")
+    eng = _engine()
+    out1 = eng.step(s)
+    F1 = _full_forces(s, out1)
+
+    core = CHGNetCore.seeded(seed=0).float()
+    model = CHGNet_Dist.from_existing(core, dtype=torch.float32)
+    model.enable_distributed_mode([0, 0])
+    pot = Potential_Dist(model, calc_forces=True)
+    E2, F2, _, _ = pot.forward(s)
+    assert abs(E2.item() - out1["energy"].item()) < 5e-3 * max(
+        1.0, abs(out1["energy"].item()))
+    assert np.abs(F2.cpu().numpy() - F1).max() < 5e-3
+
+
+@requires_gpu
+def test_verlet_reuse_matches_rebuild_at_scale():
+    """Verlet-skin reuse == fresh rebuild at 100k scale on GPU (exact
+    masking claim at size)."""
+    from distmlip_amd.structures import workload
+    s = workload("li100k")
+    eng = _engine()
+    out_fresh = eng.step(s)
+    out_verlet = eng.step_verlet(s, skin=1.0)     # first call: build+mask
+    dE = abs(out_fresh["energy"].item() - out_verlet["energy"].item())
+    dF = (out_fresh["forces_owned"] -
+          out_verlet["forces_owned"]).abs().max().item()
+    assert dE < 1e-6 * max(1.0, abs(out_fresh["energy"].item())), dE
+    assert dF < 1e-5, dF
+    # move a little, reuse, and compare against a fresh-built step
+    s2 = copy.deepcopy(s)
+    rng = np.random.default_rng(3)
+    cart = s2.frac_coords @ s2.lattice + rng.normal(0, 0.03,
+                                                    (s2.num_atoms, 3))
+    s2.frac_coords = np.mod(cart @ np.linalg.inv(s2.lattice), 1.0)
+    out_v2 = eng.step_verlet(s2, skin=1.0)
+    assert eng._vcache["rebuilds"] == 1, "reuse did not engage"
+    out_f2 = eng.step(s2)
+    dF2 = (out_f2["forces_owned"] - out_v2["forces_owned"]).abs().max().item()
+    assert dF2 < 1e-4, dF2
