@@ -461,6 +461,79 @@ __global__ void k_gated_combine_bwd(const float* __restrict__ go,
     }
 }
 
+// ---------------------------------------------------------------------------
+// GPU cell-list neighbor search (diagonal lattice, full PBC, min-image via
+// cell wrap).  fp64 math replicates the CPU builder's decisions exactly.
+// One thread per CENTER atom; center = dst so emission order is already
+// the dst-sorted scatter layout.
+// ---------------------------------------------------------------------------
+template <bool FILL>
+__global__ void k_nl(const double* __restrict__ pos,
+                     const int32_t* __restrict__ cid,
+                     const int32_t* __restrict__ order,
+                     const int32_t* __restrict__ cell_start,
+                     int32_t ncx, int32_t ncy, int32_t ncz,
+                     double lx, double ly, double lz,
+                     double r2tol, double tol, double br2tol,
+                     const int32_t* __restrict__ row_ptr,
+                     int32_t* __restrict__ cnt_or_src,
+                     int8_t* __restrict__ off_i8,
+                     uint8_t* __restrict__ bond_flag, int64_t N) {
+    for (int64_t c = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         c < N; c += (int64_t)gridDim.x * blockDim.x) {
+        const double xc = pos[3 * c], yc = pos[3 * c + 1], zc = pos[3 * c + 2];
+        const int32_t cc = cid[c];
+        const int32_t cx = cc / (ncy * ncz);
+        const int32_t cy = (cc / ncz) % ncy;
+        const int32_t cz = cc % ncz;
+        int32_t n_emit = 0;
+        int64_t w = FILL ? row_ptr[c] : 0;
+        for (int dx = -1; dx <= 1; ++dx) {
+            int32_t ax = cx + dx; int mx = 0;
+            if (ax < 0) { ax += ncx; mx = -1; }
+            else if (ax >= ncx) { ax -= ncx; mx = 1; }
+            const double sx = mx * lx;
+            for (int dy = -1; dy <= 1; ++dy) {
+                int32_t ay = cy + dy; int my = 0;
+                if (ay < 0) { ay += ncy; my = -1; }
+                else if (ay >= ncy) { ay -= ncy; my = 1; }
+                const double sy = my * ly;
+                for (int dz = -1; dz <= 1; ++dz) {
+                    int32_t az = cz + dz; int mz = 0;
+                    if (az < 0) { az += ncz; mz = -1; }
+                    else if (az >= ncz) { az -= ncz; mz = 1; }
+                    const double sz = mz * lz;
+                    const int32_t cell = (ax * ncy + ay) * ncz + az;
+                    const int32_t lo = cell_start[cell], hi = cell_start[cell + 1];
+                    for (int32_t q = lo; q < hi; ++q) {
+                        const int32_t j = order[q];
+                        if (j == c) continue;           // fpis.c:833
+                        const double ddx = pos[3 * j] + sx - xc;
+                        const double ddy = pos[3 * j + 1] + sy - yc;
+                        const double ddz = pos[3 * j + 2] + sz - zc;
+                        const double d2 = ddx * ddx + ddy * ddy + ddz * ddz;
+                        if (d2 < r2tol && d2 > tol) {
+                            if (FILL) {
+                                cnt_or_src[w] = j;
+                                // bv = pos[dst=c] + off*L - pos[src=j]
+                                // must equal -(pos_j + m*L - pos_c): off = -m
+                                off_i8[3 * w] = (int8_t)(-mx);
+                                off_i8[3 * w + 1] = (int8_t)(-my);
+                                off_i8[3 * w + 2] = (int8_t)(-mz);
+                                bond_flag[w] = (d2 < br2tol) ? 1 : 0;
+                                ++w;
+                            } else {
+                                ++n_emit;
+                            }
+                        }
+                    }
+                }
+            }
+        }
+        if (!FILL) cnt_or_src[c] = n_emit;
+    }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -629,6 +702,35 @@ int dm_silu_bwd_f32(const float* go_h, const float* go_z, const float* z,
     hipStream_t s = (hipStream_t)stream;
     k_silu_bwd<<<nblocks(total, BLOCK), BLOCK, 0, s>>>(go_h, go_z, z, dz,
                                                        total);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_nl_count_f64(const double* pos, const int32_t* cid,
+                    const int32_t* order, const int32_t* cell_start,
+                    int32_t ncx, int32_t ncy, int32_t ncz,
+                    double lx, double ly, double lz,
+                    double r2tol, double tol, int32_t* cnt, int64_t N,
+                    uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    k_nl<false><<<nblocks(N, BLOCK), BLOCK, 0, s>>>(
+        pos, cid, order, cell_start, ncx, ncy, ncz, lx, ly, lz, r2tol, tol,
+        0.0, nullptr, cnt, nullptr, nullptr, N);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_nl_fill_f64(const double* pos, const int32_t* cid,
+                   const int32_t* order, const int32_t* cell_start,
+                   int32_t ncx, int32_t ncy, int32_t ncz,
+                   double lx, double ly, double lz,
+                   double r2tol, double tol, double br2tol,
+                   const int32_t* row_ptr, int32_t* src, int8_t* off_i8,
+                   uint8_t* bond_flag, int64_t N, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    k_nl<true><<<nblocks(N, BLOCK), BLOCK, 0, s>>>(
+        pos, cid, order, cell_start, ncx, ncy, ncz, lx, ly, lz, r2tol, tol,
+        br2tol, row_ptr, src, off_i8, bond_flag, N);
     DM_CHECK_LAUNCH();
     return 0;
 }

@@ -107,6 +107,29 @@ int dm_rbf_env_bwd_f32(const float* go_exp, const float* d, const float* freqs,
                        float cutoff, int32_t pexp, int32_t nrbf, float* gd,
                        int64_t M, uint64_t stream);
 
+/* GPU-resident neighbor list (single-partition fast path; diagonal
+ * lattice, full PBC, >= 3 cells per dim).  Exact fp64 replica of the CPU
+ * builder's edge condition (fpis.c:833 contract: d^2 < r^2+tol,
+ * d^2 > tol, no self edges in any image).  Atoms are pre-binned on the
+ * host side (torch): `order` lists atom ids sorted by cell id,
+ * `cell_start[ncell+1]` the bin offsets.  Edges are emitted center-major
+ * with center as DST (so the emission order IS the dst-sorted scatter
+ * layout); `off_i8` is the integer image of the dst atom and `bond_flag`
+ * marks d^2 < bond_r^2+tol. */
+int dm_nl_count_f64(const double* pos, const int32_t* cid,
+                    const int32_t* order, const int32_t* cell_start,
+                    int32_t ncx, int32_t ncy, int32_t ncz,
+                    double lx, double ly, double lz,
+                    double r2tol, double tol, int32_t* cnt, int64_t N,
+                    uint64_t stream);
+int dm_nl_fill_f64(const double* pos, const int32_t* cid,
+                   const int32_t* order, const int32_t* cell_start,
+                   int32_t ncx, int32_t ncy, int32_t ncz,
+                   double lx, double ly, double lz,
+                   double r2tol, double tol, double br2tol,
+                   const int32_t* row_ptr, int32_t* src, int8_t* off_i8,
+                   uint8_t* bond_flag, int64_t N, uint64_t stream);
+
 const char* dm_hip_last_error(void);
 
 #ifdef __cplusplus
