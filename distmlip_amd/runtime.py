@@ -152,7 +152,8 @@ class SpmdEngine:
 
     def __init__(self, core: CHGNetCore, world: int, threads: int = 8,
                  use_bond_graph: bool = True, device: Optional[str] = None,
-                 ops=None, graph_backend=None, checkpoint: str = "auto"):
+                 ops=None, graph_backend=None, checkpoint: str = "auto",
+                 gpu_build: str = "auto"):
         self.rank = dist.get_rank() if dist.is_initialized() else 0
         assert world == 1 or dist.is_initialized(), \
             "world > 1 needs an initialized torch.distributed process group"
@@ -181,6 +182,10 @@ class SpmdEngine:
         # Halo exchanges stay OUTSIDE checkpointed segments (a re-run of a
         # comm op during backward would desynchronize the ranks).
         self.checkpoint = checkpoint
+        # GPU-resident graph build for the single-partition path
+        # (distmlip_amd/gpu_graph.py); "auto" engages on diagonal-lattice
+        # full-PBC structures when world == 1 and no custom backend is set
+        self.gpu_build = gpu_build
 
     # -- graph ------------------------------------------------------------
 
@@ -252,16 +257,35 @@ class SpmdEngine:
         cfg, core, ops, dev = self.config, self.core, self.ops, self.device
         ft = self.float_th
 
+        gpu_pd = None
         if dist_info is None:
-            dist_info = self.build_graph(structure)
+            from distmlip_amd import gpu_graph
+            if (P == 1 and self.gpu_build == "auto"
+                    and self.graph_backend is None
+                    and dev.type == "cuda"
+                    and gpu_graph.supported(structure, cfg.cutoff)):
+                gpu_pd = gpu_graph.build(
+                    structure, cfg.cutoff, cfg.three_body_cutoff, 1e-8,
+                    self.use_bond_graph, dev, frac_override=frac_override)
+            else:
+                dist_info = self.build_graph(structure)
         _HaloSeq.fwd = 0
         _HaloSeq.expect_bwd = None
-        pd = PartitionData(dist_info, r, dev, self.use_bond_graph)
-        plan = halo_plan(dist_info.markers[r], r, P)
-        line_plan = halo_plan(dist_info.line_markers[r], r, P) \
-            if self.use_bond_graph else None
-        gids = np.asarray(dist_info.global_ids[r])
-        idx_dtype = torch.long if dev.type != "cuda" else torch.long
+        if gpu_pd is not None:
+            pd = gpu_pd
+            plan, line_plan = [], []
+            gids = None                       # identity: local ids == global
+            n_owned = pd.n_atoms
+        else:
+            pd = PartitionData(dist_info, r, dev, self.use_bond_graph)
+            plan = halo_plan(dist_info.markers[r], r, P)
+            line_plan = halo_plan(dist_info.line_markers[r], r, P) \
+                if self.use_bond_graph else None
+            gids = np.asarray(dist_info.global_ids[r])
+            n_owned = dist_info.num_owned_atoms(r)
+
+        def _halo(feat, pl):
+            return feat if not pl else HaloExchange.apply(feat, pl)
 
         # ---- local geometry (per-rank; no GPU0 serialization point)
         lat0 = torch.tensor(np.asarray(structure.lattice), dtype=ft, device=dev)
@@ -270,20 +294,24 @@ class SpmdEngine:
             strain.requires_grad_(True)
         lattice = lat0 @ (torch.eye(3, device=dev, dtype=ft) + strain)
 
-        frac_src = frac_override if frac_override is not None \
-            else structure.frac_coords
-        frac_local = torch.tensor(np.asarray(frac_src)[gids],
+        frac_src = np.asarray(frac_override if frac_override is not None
+                              else structure.frac_coords)
+        frac_local = torch.tensor(frac_src if gids is None else frac_src[gids],
                                   dtype=ft, device=dev)
         pos = frac_local @ lattice
         if not pos.requires_grad:
             pos.requires_grad_(True)
         pos.retain_grad()
 
-        species_local = torch.tensor(np.asarray(structure.species)[gids],
+        spec = np.asarray(structure.species)
+        species_local = torch.tensor(spec if gids is None else spec[gids],
                                      dtype=torch.long, device=dev)
 
-        csr = dist_info.csr_parts[r] if getattr(dist_info, "csr_parts", None) else None
-        if csr is not None:
+        if gpu_pd is not None:
+            off_local = pd.off_i8.to(ft)
+            csr = None
+        elif (csr := dist_info.csr_parts[r]
+              if getattr(dist_info, "csr_parts", None) else None) is not None:
             # int8 integer images straight from the builder: 15 MB H2D at
             # 5M edges instead of a 120 MB host-side f64 gather+convert
             off_local = torch.from_numpy(csr["offsets_i8"]).to(dev).to(ft)
@@ -317,8 +345,8 @@ class SpmdEngine:
                 0, pd.map_ude, bond_dist[pd.map_de])
             nd_vec = torch.empty(pd.n_bonds, 3, dtype=ft, device=dev).index_copy(
                 0, pd.map_ude, bond_vec[pd.map_de])
-            nd_dist = HaloExchange.apply(nd_dist.unsqueeze(1), line_plan).squeeze(1)
-            nd_vec = HaloExchange.apply(nd_vec, line_plan)
+            nd_dist = _halo(nd_dist.unsqueeze(1), line_plan).squeeze(1)
+            nd_vec = _halo(nd_vec, line_plan)
 
             bond_mask = None
             if mask_cutoffs:
@@ -332,7 +360,7 @@ class SpmdEngine:
             a = core.angle_embedding(fourier_expansion(theta, core.angle_freq))
             n = torch.empty(pd.n_bonds, cfg.dim, dtype=ft, device=dev).index_copy(
                 0, pd.map_ude, e[pd.map_de])
-            n = HaloExchange.apply(n, line_plan)
+            n = _halo(n, line_plan)
 
         d = cfg.dim
         ckpt = self.checkpoint == "on" or (
@@ -381,8 +409,8 @@ class SpmdEngine:
             v, e = atom_conv(layer_i, v, e)
             if use_bg:
                 n = n.index_copy(0, pd.map_ude, e[pd.map_de])   # edge_to_bond
-                n = HaloExchange.apply(n, line_plan)
-                v = HaloExchange.apply(v, plan)
+                n = _halo(n, line_plan)
+                v = _halo(v, plan)
 
                 blk = core.bond_convs[layer_i]
 
@@ -408,7 +436,7 @@ class SpmdEngine:
                     # (a never feeds the energy after it; the reference
                     # still computes it, chgnet.py:353-368) — skip it and
                     # the halo that feeds it
-                    n = HaloExchange.apply(n, line_plan)
+                    n = _halo(n, line_plan)
 
                     def angle_body(n, a, v, _blk=blk):
                         return gated_mlp_split4(_blk.angle_mlp, n, a, v, pd,
@@ -416,15 +444,14 @@ class SpmdEngine:
 
                     a = _ck(angle_body, n, a, v)
             else:
-                v = HaloExchange.apply(v, plan)
+                v = _halo(v, plan)
 
         site_props = core.sitewise_readout(v)              # chgnet.py:391-398
 
         v, e = atom_conv(-1, v, e)                         # final atom block
-        v = HaloExchange.apply(v, plan)
+        v = _halo(v, plan)
 
         atom_e = core.final_layer(v)
-        n_owned = dist_info.num_owned_atoms(r)
         e_local_raw = atom_e[:n_owned].sum()
 
         refs_local = core.element_refs[species_local[:n_owned]].sum()
@@ -452,7 +479,9 @@ class SpmdEngine:
 
         out = {"energy": total_e, "forces_owned": forces_owned,
                "site_props_owned": site_props[:n_owned].detach(),
-               "n_owned": n_owned, "global_ids_owned": gids[:n_owned]}
+               "n_owned": n_owned,
+               "global_ids_owned": (np.arange(n_owned) if gids is None
+                                    else gids[:n_owned])}
         if calc_stresses:
             sg = gv[1].detach().clone()
             if P > 1:

@@ -232,6 +232,87 @@ def test_spmd_engine_world1_vs_oracle():
 
 
 @requires_gpu
+def test_gpu_graph_matches_cpu_builder():
+    """GPU cell-list graph build == native CPU builder: identical edge,
+    bond and line-edge SETS (fp64 decisions replicated exactly)."""
+    from distmlip_amd import capi, gpu_graph
+    from distmlip_amd.structures import bcc_li
+    s = bcc_li(8, jitter=0.12, seed=4)      # 1024 atoms, 28 A cubic
+    assert gpu_graph.supported(s, 6.0)
+    pd = gpu_graph.build(s, 6.0, 3.0, 1e-8, True, torch.device("cuda:0"))
+    cpu = capi.get_subgraphs_fast(s.cart_coords, 6.0, s.pbc, s.lattice, 1,
+                                  3.0, 1e-8, 4, True, s.frac_coords,
+                                  return_csr=True)
+    cpu_t, cpu_csr = cpu
+
+    def keys(src, dst, off):
+        src = np.asarray(src, dtype=np.int64)
+        dst = np.asarray(dst, dtype=np.int64)
+        off = np.asarray(off, dtype=np.int64)
+        return set(zip(src.tolist(), dst.tolist(), off[:, 0].tolist(),
+                       off[:, 1].tolist(), off[:, 2].tolist()))
+
+    g_src = pd.src.cpu().numpy()
+    g_dst = pd.dst.cpu().numpy()
+    g_off = pd.off_i8.cpu().numpy()
+    k_gpu = keys(g_src, g_dst, g_off)
+    k_cpu = keys(cpu_t[0][0], cpu_t[1][0], cpu_csr[0]["offsets_i8"])
+    assert k_gpu == k_cpu, (len(k_gpu), len(k_cpu),
+                            len(k_gpu ^ k_cpu))
+
+    # bond sets via the underlying edge keys
+    mde = pd.map_de.cpu().numpy()
+    b_gpu = keys(g_src[mde], g_dst[mde], g_off[mde])
+    cmde = np.asarray(cpu_t[14][0])
+    b_cpu = keys(np.asarray(cpu_t[0][0])[cmde], np.asarray(cpu_t[1][0])[cmde],
+                 cpu_csr[0]["offsets_i8"][cmde])
+    assert b_gpu == b_cpu
+
+    # line-edge sets as (edge-key(src bde), edge-key(dst bde))
+    def ek(src, dst, off, i):
+        return (int(src[i]), int(dst[i]), int(off[i, 0]), int(off[i, 1]),
+                int(off[i, 2]))
+    ls = pd.l_src.cpu().numpy()
+    ld = pd.l_dst.cpu().numpy()
+    lines_gpu = set((ek(g_src, g_dst, g_off, mde[a]),
+                     ek(g_src, g_dst, g_off, mde[b]))
+                    for a, b in zip(ls.tolist(), ld.tolist()))
+    cls = np.asarray(cpu_t[9][0])
+    cld = np.asarray(cpu_t[10][0])
+    cude = np.asarray(cpu_t[15][0])
+    inv_ude = np.empty_like(cude)
+    inv_ude[cude] = np.arange(len(cude))
+    csrc = np.asarray(cpu_t[0][0])
+    cdst = np.asarray(cpu_t[1][0])
+    coff = cpu_csr[0]["offsets_i8"]
+    lines_cpu = set((ek(csrc, cdst, coff, cmde[inv_ude[a]]),
+                     ek(csrc, cdst, coff, cmde[inv_ude[b]]))
+                    for a, b in zip(cls.tolist(), cld.tolist()))
+    assert lines_gpu == lines_cpu
+
+
+@requires_gpu
+def test_engine_gpu_build_matches_cpu_build():
+    """E+F through the GPU-built graph == through the CPU-built graph."""
+    from distmlip_amd.model import CHGNetCore
+    from distmlip_amd.runtime import SpmdEngine
+    from distmlip_amd.structures import bcc_li
+    s = bcc_li(8, jitter=0.12, seed=4)
+    core = CHGNetCore.seeded(seed=0).float()
+    e_gpu = SpmdEngine(core, world=1, threads=4, gpu_build="auto")
+    e_cpu = SpmdEngine(core, world=1, threads=4, gpu_build="off")
+    o1 = e_gpu.step(s)
+    o2 = e_cpu.step(s)
+    assert abs(o1["energy"].item() - o2["energy"].item()) < 1e-4 * max(
+        1.0, abs(o2["energy"].item()))
+    F1 = np.zeros((s.num_atoms, 3))
+    F1[o1["global_ids_owned"]] = o1["forces_owned"].cpu().numpy()
+    F2 = np.zeros((s.num_atoms, 3))
+    F2[o2["global_ids_owned"]] = o2["forces_owned"].cpu().numpy()
+    assert np.abs(F1 - F2).max() < 5e-4, np.abs(F1 - F2).max()
+
+
+@requires_gpu
 def test_native_so_loaded():
     """Guard against silent eager fallback: the HIP extension must be the
     library the process actually loaded."""
