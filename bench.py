@@ -43,7 +43,7 @@ def parse_args():
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--workload", type=str, default="li100k",
-                    choices=["li100k", "si1m", "si1k"])
+                    choices=["li100k", "si1m", "si2m", "si1k"])
     ap.add_argument("--threads", type=int, default=0,
                     help="graph-builder threads (0 = cpu_count/world)")
     ap.add_argument("--verlet", type=float, default=0.0, metavar="SKIN",

@@ -473,6 +473,15 @@ class SpmdEngine:
         # scalar reductions for reporting
         scal = torch.stack([e_local_raw.detach(), refs_local.detach()])
         if P > 1:
+            if not getattr(self, "_validated", False):
+                # one-time partition sanity: owned sets must tile the
+                # structure exactly across ranks
+                t = torch.tensor([float(n_owned)], device=dev)
+                dist.all_reduce(t)
+                assert int(round(t.item())) == structure.num_atoms, (
+                    f"owned atoms across ranks {int(t.item())} != "
+                    f"{structure.num_atoms} — rank-divergent partitioning")
+                self._validated = True
             dist.all_reduce(scal)
         total_e = core.data_std.detach() * scal[0] + core.data_mean.detach() \
             + scal[1]

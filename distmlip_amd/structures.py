@@ -126,4 +126,6 @@ def workload(name: str, n_gpus: int = 1, seed: int = 0) -> Structure:
         return bcc_li(reps, jitter=0.1, seed=seed)
     if name == "si1m":
         return diamond_si(50, jitter=0.1, seed=seed)
+    if name == "si2m":
+        return diamond_si(63, jitter=0.1, seed=seed)   # 2,000,376 atoms
     raise ValueError(f"unknown workload {name!r}")
