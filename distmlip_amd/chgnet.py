@@ -30,7 +30,6 @@ from typing import List, Optional
 import numpy as np
 import torch
 from torch import nn
-from torch.nn import functional as F
 
 import distmlip_amd
 from distmlip_amd.model import (

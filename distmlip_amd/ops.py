@@ -164,7 +164,13 @@ class _GatherAdd3(torch.autograd.Function):
         _chk_f32(zs, zd, ze)
         ctx.n_nodes = zs.shape[0]
         z = torch.empty_like(ze)
-        h = torch.empty_like(ze)
+        if torch.is_grad_enabled():
+            h = torch.empty_like(ze)
+        else:
+            # no-grad (checkpoint outer pass / inference): silu in place
+            # over z — the kernel writes z then silu(z) per element, so a
+            # single buffer is safe and halves the transient footprint
+            h = z
         _check(hip_lib().dm_gather_add3_f32(
             _fp(zs), _fp(zd), _fp(ze), _ip(src), _ip(dst), _fp(z), _fp(h),
             ze.shape[0], ze.shape[1], _stream()), "dm_gather_add3_f32")
@@ -195,7 +201,7 @@ class _GatherAdd4(torch.autograd.Function):
         ctx.n_bonds = z1.shape[0]
         ctx.n_nodes = zv.shape[0]
         z = torch.empty_like(za)
-        h = torch.empty_like(za)
+        h = torch.empty_like(za) if torch.is_grad_enabled() else z
         _check(hip_lib().dm_gather_add4_f32(
             _fp(z1), _fp(z2), _fp(za), _fp(zv), _ip(pd.l_src), _ip(pd.l_dst),
             _ip(pd.center), _fp(z), _fp(h), za.shape[0], za.shape[1],

@@ -14,12 +14,15 @@ the permutation CSRs the graph builder emits; `csr` is an optional
 (perm, row_ptr) pair giving the deterministic backward of a gather.
 
     gather(x, idx, csr)                  -> x[idx]
-    gather_add3(zs, zd, ze, pd)          -> zs[pd.src] + zd[pd.dst] + ze
-    gather_add4(z1, z2, za, zv, pd)      -> z1[pd.l_src] + z2[pd.l_dst]
-                                            + za + zv[pd.center]
+    gather_add3_act(zs, zd, ze, pd)      -> silu(zs[pd.src] + zd[pd.dst] + ze)
+    gather_add4_act(z1, z2, za, zv, pd)  -> silu(z1[pd.l_src] + z2[pd.l_dst]
+                                                 + za + zv[pd.center])
     scatter_edges(msg, pd, base)         -> base + segment-sum of the
                                             dst-sorted msg rows per node
     scatter_lines(msg, pd, base)         -> same over the line CSR per bond
+    gated_combine(c, g, w, base)         -> base + silu(c)*sigmoid(g)*w
+    edge_geom_rbf(pos, off, freqs, c, p, pd) -> (bond_vec, bond_dist, rbf*env)
+    rbf_env(d, freqs, cutoff, pexp)      -> rbf*env
 """
 from __future__ import annotations
 
@@ -33,9 +36,15 @@ class OpsBackend(Protocol):
                csr: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
                ) -> torch.Tensor: ...
 
-    def gather_add3(self, zs, zd, ze, pd) -> torch.Tensor: ...
+    def gather_add3_act(self, zs, zd, ze, pd) -> torch.Tensor: ...
 
-    def gather_add4(self, z1, z2, za, zv, pd) -> torch.Tensor: ...
+    def gather_add4_act(self, z1, z2, za, zv, pd) -> torch.Tensor: ...
+
+    def gated_combine(self, c, g, w=None, base=None) -> torch.Tensor: ...
+
+    def edge_geom_rbf(self, pos, offshift, freqs, cutoff, pexp, pd): ...
+
+    def rbf_env(self, d, freqs, cutoff, pexp) -> torch.Tensor: ...
 
     def scatter_edges(self, msg, pd, base=None) -> torch.Tensor: ...
 
