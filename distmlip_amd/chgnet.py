@@ -62,6 +62,16 @@ def _cg_weight(mlp: GatedMLP):
     return w, b
 
 
+def _second_layer_packed(mlp: GatedMLP, h, d: int):
+    """core2|gate2 as ONE batched GEMM over the packed hidden halves:
+    h [*,2d] viewed as [2,*,d] -> cg [2,*,d]; no slicing, so no
+    slice-gradient zero/copy/add passes in backward."""
+    hb = h.view(-1, 2, d).transpose(0, 1)
+    w2 = torch.stack([mlp.core2.weight.t(), mlp.gate2.weight.t()])
+    b2 = torch.stack([mlp.core2.bias.unsqueeze(0), mlp.gate2.bias.unsqueeze(0)])
+    return torch.baddbmm(b2, hb, w2)
+
+
 def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
     """GatedMLP over cat(v[src], v[dst], e) via split-linear + one 2h-wide
     gather_add3, finished by the fused gated-combine epilogue:
@@ -69,9 +79,8 @@ def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
     wcg, bcg = _cg_weight(mlp)
     ws, wd, we = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:]
     h = ops.gather_add3_act(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
-    c = mlp.core2(h[:, :d])
-    g = mlp.gate2(h[:, d:])
-    return ops.gated_combine(c, g, w, base)
+    cg = _second_layer_packed(mlp, h, d)
+    return ops.gated_combine_packed(cg, w, base)
 
 
 def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
@@ -82,9 +91,8 @@ def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
                       wcg[:, 3 * d:])
     h = ops.gather_add4_act(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
                             v @ wv.t(), pd)
-    c = mlp.core2(h[:, :d])
-    g = mlp.gate2(h[:, d:])
-    return ops.gated_combine(c, g, w, base)
+    cg = _second_layer_packed(mlp, h, d)
+    return ops.gated_combine_packed(cg, w, base)
 
 
 class PartitionData:

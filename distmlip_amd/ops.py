@@ -164,12 +164,14 @@ class _GatherAdd3(torch.autograd.Function):
         _chk_f32(zs, zd, ze)
         ctx.n_nodes = zs.shape[0]
         z = torch.empty_like(ze)
-        if torch.is_grad_enabled():
+        # grad mode is always OFF inside Function.forward, so test the
+        # inputs: when none requires grad (checkpoint outer pass /
+        # inference) no backward will run and silu can go in place over z —
+        # the kernel writes z then silu(z) per element, so a single buffer
+        # is safe and halves the transient footprint
+        if zs.requires_grad or zd.requires_grad or ze.requires_grad:
             h = torch.empty_like(ze)
         else:
-            # no-grad (checkpoint outer pass / inference): silu in place
-            # over z — the kernel writes z then silu(z) per element, so a
-            # single buffer is safe and halves the transient footprint
             h = z
         _check(hip_lib().dm_gather_add3_f32(
             _fp(zs), _fp(zd), _fp(ze), _ip(src), _ip(dst), _fp(z), _fp(h),
@@ -201,7 +203,10 @@ class _GatherAdd4(torch.autograd.Function):
         ctx.n_bonds = z1.shape[0]
         ctx.n_nodes = zv.shape[0]
         z = torch.empty_like(za)
-        h = torch.empty_like(za) if torch.is_grad_enabled() else z
+        # see _GatherAdd3.forward: in-place silu only when no backward runs
+        needs_h = (z1.requires_grad or z2.requires_grad or za.requires_grad
+                   or zv.requires_grad)
+        h = torch.empty_like(za) if needs_h else z
         _check(hip_lib().dm_gather_add4_f32(
             _fp(z1), _fp(z2), _fp(za), _fp(zv), _ip(pd.l_src), _ip(pd.l_dst),
             _ip(pd.center), _fp(z), _fp(h), za.shape[0], za.shape[1],
@@ -343,6 +348,49 @@ class _RbfEnv(torch.autograd.Function):
         return gd, None, None, None
 
 
+class _GatedCombinePacked(torch.autograd.Function):
+    """gated_combine over PACKED cg [2,E,D] (cg[0]=c, cg[1]=g): avoids the
+    h-slice / c-g-select gradient zero+copy+add passes entirely."""
+
+    @staticmethod
+    def forward(ctx, cg, w, base):
+        _chk_f32(cg, w, base)
+        half = cg.shape[1] * cg.shape[2]
+        c_ptr = ctypes.cast(cg.data_ptr(), POINTER(c_float))
+        g_ptr = ctypes.cast(cg.data_ptr() + 4 * half, POINTER(c_float))
+        out = torch.empty(cg.shape[1], cg.shape[2], dtype=cg.dtype,
+                          device=cg.device)
+        _check(hip_lib().dm_gated_combine_fwd_f32(
+            c_ptr, g_ptr, _fp(w) if w is not None else None,
+            _fp(base) if base is not None else None, _fp(out), half,
+            _stream()), "dm_gated_combine_fwd_f32")
+        ctx.save_for_backward(cg, w) if w is not None else             ctx.save_for_backward(cg)
+        ctx.has_w = w is not None
+        ctx.has_base = base is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        if ctx.has_w:
+            cg, w = ctx.saved_tensors
+        else:
+            (cg,) = ctx.saved_tensors
+            w = None
+        go = go.contiguous()
+        half = cg.shape[1] * cg.shape[2]
+        dcg = torch.empty_like(cg)
+        c_ptr = ctypes.cast(cg.data_ptr(), POINTER(c_float))
+        g_ptr = ctypes.cast(cg.data_ptr() + 4 * half, POINTER(c_float))
+        dc_ptr = ctypes.cast(dcg.data_ptr(), POINTER(c_float))
+        dg_ptr = ctypes.cast(dcg.data_ptr() + 4 * half, POINTER(c_float))
+        dw = torch.empty_like(w) if ctx.has_w else None
+        _check(hip_lib().dm_gated_combine_bwd_f32(
+            _fp(go), c_ptr, g_ptr, _fp(w) if w is not None else None,
+            dc_ptr, dg_ptr, _fp(dw) if dw is not None else None, half,
+            _stream()), "dm_gated_combine_bwd_f32")
+        return dcg, dw, (go if ctx.has_base else None)
+
+
 class HipOps:
     """Product ops backend (see ops_base.OpsBackend)."""
 
@@ -375,6 +423,12 @@ class HipOps:
     def gated_combine(self, c, g, w=None, base=None):
         return _GatedCombine.apply(
             c.contiguous(), g.contiguous(),
+            w.contiguous() if w is not None else None,
+            base.contiguous() if base is not None else None)
+
+    def gated_combine_packed(self, cg, w=None, base=None):
+        return _GatedCombinePacked.apply(
+            cg.contiguous(),
             w.contiguous() if w is not None else None,
             base.contiguous() if base is not None else None)
 

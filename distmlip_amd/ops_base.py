@@ -21,6 +21,10 @@ the permutation CSRs the graph builder emits; `csr` is an optional
                                             dst-sorted msg rows per node
     scatter_lines(msg, pd, base)         -> same over the line CSR per bond
     gated_combine(c, g, w, base)         -> base + silu(c)*sigmoid(g)*w
+    gated_combine_packed(cg, w, base)    -> same over PACKED cg [2,*,D]
+                                            (cg[0]=c, cg[1]=g; lets the
+                                            second-layer core|gate GEMMs run
+                                            as one bmm with no slicing)
     edge_geom_rbf(pos, off, freqs, c, p, pd) -> (bond_vec, bond_dist, rbf*env)
     rbf_env(d, freqs, cutoff, pexp)      -> rbf*env
 """
@@ -41,6 +45,8 @@ class OpsBackend(Protocol):
     def gather_add4_act(self, z1, z2, za, zv, pd) -> torch.Tensor: ...
 
     def gated_combine(self, c, g, w=None, base=None) -> torch.Tensor: ...
+
+    def gated_combine_packed(self, cg, w=None, base=None) -> torch.Tensor: ...
 
     def edge_geom_rbf(self, pos, offshift, freqs, cutoff, pexp, pd): ...
 

@@ -217,6 +217,9 @@ class CpuRefOps:
             out = out * w
         return out if base is None else base + out
 
+    def gated_combine_packed(self, cg, w=None, base=None):
+        return self.gated_combine(cg[0], cg[1], w, base)
+
     def edge_geom_rbf(self, pos, offshift, freqs, cutoff, pexp, pd):
         from distmlip_amd.model import bond_expansion_from_dist
         bv = pos[pd.dst] + offshift - pos[pd.src]

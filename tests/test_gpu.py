@@ -176,6 +176,69 @@ def test_gated_combine_fwd_bwd():
             assert torch.allclose(base.grad, b2.grad)
 
 
+def test_gated_combine_packed_fwd_bwd():
+    """Packed cg [2,E,D] variant matches the separate-tensor op + grads."""
+    from distmlip_amd.ops import _GatedCombinePacked
+    torch.manual_seed(2)
+    dev = torch.device("cuda:0")
+    E, D = 4000, 64
+    for has_w, has_base in [(True, True), (False, False)]:
+        cg = torch.randn(2, E, D, device=dev, requires_grad=True)
+        w = torch.randn(E, D, device=dev, requires_grad=True) if has_w else None
+        base = torch.randn(E, D, device=dev, requires_grad=True) if has_base else None
+        out = _GatedCombinePacked.apply(cg, w, base)
+        cg2 = cg.detach().clone().requires_grad_(True)
+        r = torch.nn.functional.silu(cg2[0]) * torch.sigmoid(cg2[1])
+        if has_w:
+            w2 = w.detach().clone().requires_grad_(True)
+            r = r * w2
+        if has_base:
+            b2 = base.detach().clone().requires_grad_(True)
+            r = b2 + r
+        assert torch.allclose(out, r, atol=2e-6), (has_w, has_base)
+        go = torch.randn_like(out)
+        out.backward(go)
+        r.backward(go)
+        assert torch.allclose(cg.grad, cg2.grad, atol=2e-5)
+        if has_w:
+            assert torch.allclose(w.grad, w2.grad, atol=2e-5)
+        if has_base:
+            assert torch.allclose(base.grad, b2.grad)
+
+
+def test_gather_add3_inplace_only_without_grad():
+    """Regression: grad mode is off inside Function.forward, so the in-place
+    no-grad silu must key on input.requires_grad — with grad inputs, z must
+    stay the PRE-activation sum (run 15 caught silu written over z)."""
+    from distmlip_amd.ops import _GatherAdd3
+    torch.manual_seed(3)
+    dev = torch.device("cuda:0")
+    N, E, D = 300, 2000, 64
+    src64 = torch.randint(0, N, (E,))
+    dst64 = torch.sort(torch.randint(0, N, (E,))).values
+    import numpy as np
+    rp = torch.from_numpy(np.searchsorted(
+        dst64.numpy(), np.arange(N + 1)).astype(np.int32)).to(dev)
+    sp_order = torch.argsort(src64, stable=True)
+    sp = sp_order.to(torch.int32).to(dev)
+    srp = torch.from_numpy(np.searchsorted(
+        src64[sp_order].numpy(), np.arange(N + 1)).astype(np.int32)).to(dev)
+    zs = torch.randn(N, D, device=dev, requires_grad=True)
+    zd = torch.randn(N, D, device=dev)
+    ze = torch.randn(E, D, device=dev)
+    src = src64.to(torch.int32).to(dev)
+    dst = dst64.to(torch.int32).to(dev)
+    z, h = _GatherAdd3.apply(zs, zd, ze, src, dst, sp, srp, rp)
+    ref = zs.detach()[src64] + zd[dst64] + ze
+    assert torch.allclose(z, ref, atol=1e-5)          # pre-activation kept
+    assert torch.allclose(h, torch.nn.functional.silu(ref), atol=1e-5)
+    assert z.data_ptr() != h.data_ptr()
+    # and with NO grad inputs the two alias (the memory optimisation)
+    z2, h2 = _GatherAdd3.apply(zs.detach(), zd, ze, src, dst, sp, srp, rp)
+    assert z2.data_ptr() == h2.data_ptr()
+    assert torch.allclose(h2, torch.nn.functional.silu(ref), atol=1e-5)
+
+
 def _gpu_model(core, P, devices=None):
     from distmlip_amd.chgnet import CHGNet_Dist
     model = CHGNet_Dist.from_existing(core, dtype=torch.float32)
