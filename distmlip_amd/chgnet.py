@@ -54,21 +54,36 @@ def _split_first(lin: nn.Linear, parts: List[int]):
     return ws, lin.bias
 
 
-def _cg_weight(mlp: GatedMLP):
-    """[core1 ; gate1] stacked weight [2h, in] + bias [2h] (one GEMM feeds
-    both branches; the input is read once)."""
-    w = torch.cat([mlp.core1.weight, mlp.gate1.weight], dim=0)
-    b = torch.cat([mlp.core1.bias, mlp.gate1.bias], dim=0)
-    return w, b
+def _packed_weights(mlp: GatedMLP):
+    """Per-MLP packed weights, cached on the module when params are frozen
+    (they are in SpmdEngine/CHGNet_Dist inference) so the cat/stack/
+    transpose launches run once, not every step:
 
-
-def _second_layer_packed(mlp: GatedMLP, h, d: int):
-    """core2|gate2 as ONE batched GEMM over the packed hidden halves:
-    h [*,2d] viewed as [2,*,d] -> cg [2,*,d]; no slicing, so no
-    slice-gradient zero/copy/add passes in backward."""
-    hb = h.view(-1, 2, d).transpose(0, 1)
+      wcg [2h, in], bcg [2h] — [core1 ; gate1] stacked first layer (one
+      GEMM feeds both branches; the input is read once);
+      w2 [2, d, d], b2 [2, 1, d] — core2|gate2 as a batched GEMM over the
+      packed hidden halves.
+    """
+    frozen = not any(p.requires_grad for p in mlp.parameters())
+    key = (mlp.core1.weight.device, mlp.core1.weight.dtype)
+    if frozen:
+        cached = getattr(mlp, "_dm_packed", None)
+        if cached is not None and cached[0] == key:
+            return cached[1]
+    wcg = torch.cat([mlp.core1.weight, mlp.gate1.weight], dim=0)
+    bcg = torch.cat([mlp.core1.bias, mlp.gate1.bias], dim=0)
     w2 = torch.stack([mlp.core2.weight.t(), mlp.gate2.weight.t()])
     b2 = torch.stack([mlp.core2.bias.unsqueeze(0), mlp.gate2.bias.unsqueeze(0)])
+    packed = (wcg, bcg, w2.contiguous(), b2)
+    if frozen:
+        mlp._dm_packed = (key, packed)
+    return packed
+
+
+def _second_layer_packed(h, w2, b2, d: int):
+    """cg [2,*,d] = baddbmm over h [*,2d] viewed [2,*,d]; no slicing, so no
+    slice-gradient zero/copy/add passes in backward."""
+    hb = h.view(-1, 2, d).transpose(0, 1)
     return torch.baddbmm(b2, hb, w2)
 
 
@@ -76,22 +91,22 @@ def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
     """GatedMLP over cat(v[src], v[dst], e) via split-linear + one 2h-wide
     gather_add3, finished by the fused gated-combine epilogue:
     returns base + silu(core2(silu(z_c))) * sigmoid(gate2(silu(z_g))) * w."""
-    wcg, bcg = _cg_weight(mlp)
+    wcg, bcg, w2, b2 = _packed_weights(mlp)
     ws, wd, we = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:]
     h = ops.gather_add3_act(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
-    cg = _second_layer_packed(mlp, h, d)
+    cg = _second_layer_packed(h, w2, b2, d)
     return ops.gated_combine_packed(cg, w, base)
 
 
 def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
                      base=None):
     """GatedMLP over cat(n[l_src], n[l_dst], a, v[center]), same structure."""
-    wcg, bcg = _cg_weight(mlp)
+    wcg, bcg, w2p, b2p = _packed_weights(mlp)
     w1, w2, wa, wv = (wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:3 * d],
                       wcg[:, 3 * d:])
     h = ops.gather_add4_act(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
                             v @ wv.t(), pd)
-    cg = _second_layer_packed(mlp, h, d)
+    cg = _second_layer_packed(h, w2p, b2p, d)
     return ops.gated_combine_packed(cg, w, base)
 
 
