@@ -162,6 +162,11 @@ class _GatherAdd3(torch.autograd.Function):
     @staticmethod
     def forward(ctx, zs, zd, ze, src, dst, src_perm, src_row_ptr, row_ptr):
         _chk_f32(zs, zd, ze)
+        # the z output is usually discarded (only h flows on): without this,
+        # autograd materializes a full [E,2d] ZERO go_z every backward (a
+        # zero-fill write + an extra read inside silu_bwd, ~0.8 ms/call at
+        # li100k, rocprof run 17)
+        ctx.set_materialize_grads(False)
         ctx.n_nodes = zs.shape[0]
         z = torch.empty_like(ze)
         # grad mode is always OFF inside Function.forward, so test the
@@ -199,6 +204,7 @@ class _GatherAdd4(torch.autograd.Function):
     @staticmethod
     def forward(ctx, z1, z2, za, zv, pd):
         _chk_f32(z1, z2, za, zv)
+        ctx.set_materialize_grads(False)   # see _GatherAdd3.forward
         ctx.pd = pd
         ctx.n_bonds = z1.shape[0]
         ctx.n_nodes = zv.shape[0]
@@ -301,6 +307,7 @@ class _EdgeGeomRbf(torch.autograd.Function):
             float(cutoff), int(pexp), nrbf, _fp(bv), _fp(bd), _fp(exp_out),
             E, _stream()), "dm_edge_geom_rbf_fwd_f32")
         ctx.save_for_backward(bv, bd, freqs)
+        ctx.set_materialize_grads(False)   # bv/bd grads are often absent
         ctx.pd = pd
         ctx.cutoff, ctx.pexp, ctx.nrbf = float(cutoff), int(pexp), nrbf
         ctx.n_nodes = pos.shape[0]

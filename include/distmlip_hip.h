@@ -69,7 +69,10 @@ int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
 
 /* out = (base?base:0) + silu(c) * sigmoid(g) * (w?w:1) — the gated-MLP
  * epilogue (core activation x gate x shared message weight x residual),
- * fused from ~5 eager passes.  bwd emits dc, dg and (if w) dw. */
+ * fused from ~5 eager passes.  bwd emits dc, dg and (if w) dw.
+ * c and g (and dc/dg) are plain element pointers, so a PACKED [2, N, D]
+ * buffer works by passing base and base + N*D — the binding's
+ * gated_combine_packed path relies on this. */
 int dm_gated_combine_fwd_f32(const float* c, const float* g, const float* w,
                              const float* base, float* out, int64_t total,
                              uint64_t stream);
