@@ -62,7 +62,9 @@ def _packed_weights(mlp: GatedMLP):
       wcg [2h, in], bcg [2h] — [core1 ; gate1] stacked first layer (one
       GEMM feeds both branches; the input is read once);
       w2 [2, d, d], b2 [2, 1, d] — core2|gate2 as a batched GEMM over the
-      packed hidden halves.
+      packed hidden halves;
+      fusedT [d, 2h] — the per-edge/per-line input block's weight
+      (columns 2d:3d of wcg) transposed for the fused edge-MLP kernel.
     """
     frozen = not any(p.requires_grad for p in mlp.parameters())
     key = (mlp.core1.weight.device, mlp.core1.weight.dtype)
@@ -74,7 +76,9 @@ def _packed_weights(mlp: GatedMLP):
     bcg = torch.cat([mlp.core1.bias, mlp.gate1.bias], dim=0)
     w2 = torch.stack([mlp.core2.weight.t(), mlp.gate2.weight.t()])
     b2 = torch.stack([mlp.core2.bias.unsqueeze(0), mlp.gate2.bias.unsqueeze(0)])
-    packed = (wcg, bcg, w2.contiguous(), b2)
+    d = mlp.core2.weight.shape[0]
+    fusedT = wcg[:, 2 * d:3 * d].t().contiguous()
+    packed = (wcg, bcg, w2.contiguous(), b2, fusedT)
     if frozen:
         mlp._dm_packed = (key, packed)
     return packed
@@ -91,9 +95,15 @@ def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
     """GatedMLP over cat(v[src], v[dst], e) via split-linear + one 2h-wide
     gather_add3, finished by the fused gated-combine epilogue:
     returns base + silu(core2(silu(z_c))) * sigmoid(gate2(silu(z_g))) * w."""
-    wcg, bcg, w2, b2 = _packed_weights(mlp)
-    ws, wd, we = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:]
-    h = ops.gather_add3_act(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
+    wcg, bcg, w2, b2, fusedT = _packed_weights(mlp)
+    ws, wd = wcg[:, :d], wcg[:, d:2 * d]
+    if (d == 64 and not wcg.requires_grad
+            and hasattr(ops, "edge_mlp3_act")):
+        # per-edge GEMM fused into the gather kernel (no [E,2d] transient)
+        h = ops.edge_mlp3_act(e, fusedT, bcg, v @ ws.t(), v @ wd.t(), pd)
+    else:
+        we = wcg[:, 2 * d:]
+        h = ops.gather_add3_act(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
     cg = _second_layer_packed(h, w2, b2, d)
     return ops.gated_combine_packed(cg, w, base)
 
@@ -101,11 +111,16 @@ def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
 def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
                      base=None):
     """GatedMLP over cat(n[l_src], n[l_dst], a, v[center]), same structure."""
-    wcg, bcg, w2p, b2p = _packed_weights(mlp)
-    w1, w2, wa, wv = (wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:3 * d],
-                      wcg[:, 3 * d:])
-    h = ops.gather_add4_act(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
-                            v @ wv.t(), pd)
+    wcg, bcg, w2p, b2p, fusedT = _packed_weights(mlp)
+    w1, w2, wv = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 3 * d:]
+    if (d == 64 and not wcg.requires_grad
+            and hasattr(ops, "edge_mlp4_act")):
+        h = ops.edge_mlp4_act(a, fusedT, bcg, n @ w1.t(), n @ w2.t(),
+                              v @ wv.t(), pd)
+    else:
+        wa = wcg[:, 2 * d:3 * d]
+        h = ops.gather_add4_act(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
+                                v @ wv.t(), pd)
     cg = _second_layer_packed(h, w2p, b2p, d)
     return ops.gated_combine_packed(cg, w, base)
 

@@ -36,6 +36,13 @@ def hip_lib() -> ctypes.CDLL:
         lib.dm_gather_add4_f32.restype = c_int32
         lib.dm_gather_add4_f32.argtypes = [fp, fp, fp, fp, ip, ip, ip, fp,
                                            fp, c_int64, c_int64, c_uint64]
+        lib.dm_edge_mlp3_f32.restype = c_int32
+        lib.dm_edge_mlp3_f32.argtypes = [fp, fp, fp, fp, fp, ip, ip, fp, fp,
+                                         c_int64, c_int64, c_int64, c_uint64]
+        lib.dm_edge_mlp4_f32.restype = c_int32
+        lib.dm_edge_mlp4_f32.argtypes = [fp, fp, fp, fp, fp, fp, ip, ip, ip,
+                                         fp, fp, c_int64, c_int64, c_int64,
+                                         c_uint64]
         lib.dm_silu_bwd_f32.restype = c_int32
         lib.dm_silu_bwd_f32.argtypes = [fp, fp, fp, fp, c_int64, c_uint64]
         lib.dm_seg_sum_f32.restype = c_int32
@@ -196,6 +203,89 @@ class _GatherAdd3(torch.autograd.Function):
         gzs = raw_seg_sum_gather(dz, src_perm, src_row_ptr, ctx.n_nodes)
         gzd = raw_seg_sum(dz, row_ptr, ctx.n_nodes)
         return gzs, gzd, dz, None, None, None, None, None
+
+
+class _EdgeMlp3(torch.autograd.Function):
+    """Fused first-layer edge MLP over the atom graph:
+    (z, silu(z)) with z = erow @ WT + bias + zs[src] + zd[dst], the per-edge
+    GEMM computed INSIDE the gather kernel (LDS-resident WT) so the [E,2d]
+    GEMM output is never materialized.  WT/bias must be frozen (inference
+    path) — their grads are not produced."""
+
+    @staticmethod
+    def forward(ctx, erow, wt, bias, zs, zd, src, dst, src_perm, src_row_ptr,
+                row_ptr):
+        _chk_f32(erow, wt, bias, zs, zd)
+        assert not wt.requires_grad and not bias.requires_grad
+        ctx.set_materialize_grads(False)
+        ctx.n_nodes = zs.shape[0]
+        E, dout = erow.shape[0], wt.shape[1]
+        z = torch.empty(E, dout, dtype=erow.dtype, device=erow.device)
+        needs_h = erow.requires_grad or zs.requires_grad or zd.requires_grad
+        h = torch.empty_like(z) if needs_h else z
+        _check(hip_lib().dm_edge_mlp3_f32(
+            _fp(erow), _fp(wt), _fp(bias), _fp(zs), _fp(zd), _ip(src),
+            _ip(dst), _fp(z), _fp(h), E, wt.shape[0], dout, _stream()),
+            "dm_edge_mlp3_f32")
+        ctx.save_for_backward(src_perm, src_row_ptr, row_ptr, z, wt)
+        return z, h
+
+    @staticmethod
+    def backward(ctx, go_z, go_h):
+        src_perm, src_row_ptr, row_ptr, z, wt = ctx.saved_tensors
+        if go_h is not None:
+            dz = raw_silu_bwd(go_h.contiguous(),
+                              go_z.contiguous() if go_z is not None else None,
+                              z)
+        else:
+            dz = go_z.contiguous()
+        de = dz @ wt.t()
+        gzs = raw_seg_sum_gather(dz, src_perm, src_row_ptr, ctx.n_nodes)
+        gzd = raw_seg_sum(dz, row_ptr, ctx.n_nodes)
+        return de, None, None, gzs, gzd, None, None, None, None, None
+
+
+class _EdgeMlp4(torch.autograd.Function):
+    """Fused first-layer bond MLP over the line graph (3-gather form):
+    z = arow @ WT + bias + z1[l_src] + z2[l_dst] + zv[center]."""
+
+    @staticmethod
+    def forward(ctx, arow, wt, bias, z1, z2, zv, pd):
+        _chk_f32(arow, wt, bias, z1, z2, zv)
+        assert not wt.requires_grad and not bias.requires_grad
+        ctx.set_materialize_grads(False)
+        ctx.pd = pd
+        ctx.n_bonds = z1.shape[0]
+        ctx.n_nodes = zv.shape[0]
+        L, dout = arow.shape[0], wt.shape[1]
+        z = torch.empty(L, dout, dtype=arow.dtype, device=arow.device)
+        needs_h = (arow.requires_grad or z1.requires_grad or z2.requires_grad
+                   or zv.requires_grad)
+        h = torch.empty_like(z) if needs_h else z
+        _check(hip_lib().dm_edge_mlp4_f32(
+            _fp(arow), _fp(wt), _fp(bias), _fp(z1), _fp(z2), _fp(zv),
+            _ip(pd.l_src), _ip(pd.l_dst), _ip(pd.center), _fp(z), _fp(h), L,
+            wt.shape[0], dout, _stream()), "dm_edge_mlp4_f32")
+        ctx.save_for_backward(z, wt)
+        return z, h
+
+    @staticmethod
+    def backward(ctx, go_z, go_h):
+        pd = ctx.pd
+        z, wt = ctx.saved_tensors
+        if go_h is not None:
+            dz = raw_silu_bwd(go_h.contiguous(),
+                              go_z.contiguous() if go_z is not None else None,
+                              z)
+        else:
+            dz = go_z.contiguous()
+        da = dz @ wt.t()
+        gz1 = raw_seg_sum_gather(dz, pd.line_src_perm, pd.line_src_row_ptr,
+                                 ctx.n_bonds)
+        gz2 = raw_seg_sum(dz, pd.line_row_ptr, ctx.n_bonds)
+        gzv = raw_seg_sum_gather(dz, pd.center_perm, pd.center_row_ptr,
+                                 ctx.n_nodes)
+        return da, None, None, gz1, gz2, gzv, None
 
 
 class _GatherAdd4(torch.autograd.Function):
@@ -414,6 +504,19 @@ class HipOps:
         _z, h = _GatherAdd3.apply(zs.contiguous(), zd.contiguous(),
                                   ze.contiguous(), pd.src, pd.dst,
                                   pd.src_perm, pd.src_row_ptr, pd.row_ptr)
+        return h
+
+    def edge_mlp3_act(self, erow, wt, bias, zs, zd, pd):
+        """silu(erow @ wt + bias + zs[src] + zd[dst]) fused in one kernel
+        (wt = first-layer weight.T, [64,128] only)."""
+        _z, h = _EdgeMlp3.apply(erow.contiguous(), wt, bias, zs.contiguous(),
+                                zd.contiguous(), pd.src, pd.dst, pd.src_perm,
+                                pd.src_row_ptr, pd.row_ptr)
+        return h
+
+    def edge_mlp4_act(self, arow, wt, bias, z1, z2, zv, pd):
+        _z, h = _EdgeMlp4.apply(arow.contiguous(), wt, bias, z1.contiguous(),
+                                z2.contiguous(), zv.contiguous(), pd)
         return h
 
     def gather_add4_act(self, z1, z2, za, zv, pd):

@@ -17,6 +17,13 @@ the permutation CSRs the graph builder emits; `csr` is an optional
     gather_add3_act(zs, zd, ze, pd)      -> silu(zs[pd.src] + zd[pd.dst] + ze)
     gather_add4_act(z1, z2, za, zv, pd)  -> silu(z1[pd.l_src] + z2[pd.l_dst]
                                                  + za + zv[pd.center])
+    edge_mlp3_act(erow, wt, bias, zs, zd, pd)
+                                         -> silu(erow @ wt + bias
+                                                 + zs[pd.src] + zd[pd.dst])
+                                            (per-edge GEMM fused into the
+                                            gather kernel; frozen wt/bias)
+    edge_mlp4_act(arow, wt, bias, z1, z2, zv, pd)
+                                         -> the 3-gather line-graph form
     scatter_edges(msg, pd, base)         -> base + segment-sum of the
                                             dst-sorted msg rows per node
     scatter_lines(msg, pd, base)         -> same over the line CSR per bond
@@ -43,6 +50,10 @@ class OpsBackend(Protocol):
     def gather_add3_act(self, zs, zd, ze, pd) -> torch.Tensor: ...
 
     def gather_add4_act(self, z1, z2, za, zv, pd) -> torch.Tensor: ...
+
+    def edge_mlp3_act(self, erow, wt, bias, zs, zd, pd) -> torch.Tensor: ...
+
+    def edge_mlp4_act(self, arow, wt, bias, z1, z2, zv, pd) -> torch.Tensor: ...
 
     def gated_combine(self, c, g, w=None, base=None) -> torch.Tensor: ...
 

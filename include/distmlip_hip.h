@@ -67,6 +67,23 @@ int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
                           const int32_t* row_ptr, const float* base,
                           float* out, int64_t N, int64_t D, uint64_t stream);
 
+/* Fused first-layer edge MLP: z[e,:] = erow[e,:] @ WT + bias
+ *   + zs[src[e],:] + zd[dst[e],:]   (+ zv[center[e],:] for the 4-input
+ * line-graph form), out_act = silu(z).  WT is weight.T, [Din, Dout]
+ * row-major, LDS-resident; compiled for Din=64, Dout=128 (the CHGNet
+ * gated-MLP first layer) — other shapes return an error and the binding
+ * falls back to the unfused GEMM + dm_gather_add{3,4} path.  out and
+ * out_act may alias (z then silu(z) is written per element). */
+int dm_edge_mlp3_f32(const float* erow, const float* WT, const float* bias,
+                     const float* zs, const float* zd, const int32_t* src,
+                     const int32_t* dst, float* out, float* out_act,
+                     int64_t E, int64_t Din, int64_t Dout, uint64_t stream);
+int dm_edge_mlp4_f32(const float* arow, const float* WT, const float* bias,
+                     const float* z1, const float* z2, const float* zv,
+                     const int32_t* lsrc, const int32_t* ldst,
+                     const int32_t* center, float* out, float* out_act,
+                     int64_t L, int64_t Din, int64_t Dout, uint64_t stream);
+
 /* out = (base?base:0) + silu(c) * sigmoid(g) * (w?w:1) — the gated-MLP
  * epilogue (core activation x gate x shared message weight x residual),
  * fused from ~5 eager passes.  bwd emits dc, dg and (if w) dw.

@@ -220,6 +220,14 @@ class CpuRefOps:
     def gated_combine_packed(self, cg, w=None, base=None):
         return self.gated_combine(cg[0], cg[1], w, base)
 
+    def edge_mlp3_act(self, erow, wt, bias, zs, zd, pd):
+        return torch.nn.functional.silu(
+            erow @ wt + bias + zs[pd.src] + zd[pd.dst])
+
+    def edge_mlp4_act(self, arow, wt, bias, z1, z2, zv, pd):
+        return torch.nn.functional.silu(
+            arow @ wt + bias + z1[pd.l_src] + z2[pd.l_dst] + zv[pd.center])
+
     def edge_geom_rbf(self, pos, offshift, freqs, cutoff, pexp, pd):
         from distmlip_amd.model import bond_expansion_from_dist
         bv = pos[pd.dst] + offshift - pos[pd.src]
