@@ -85,9 +85,12 @@ def _packed_weights(mlp: GatedMLP):
 
 
 def _second_layer_packed(h, w2, b2, d: int):
-    """cg [2,*,d] = baddbmm over h [*,2d] viewed [2,*,d]; no slicing, so no
-    slice-gradient zero/copy/add passes in backward."""
-    hb = h.view(-1, 2, d).transpose(0, 1)
+    """cg [2,*,d] = baddbmm over the packed hidden halves; no slicing, so
+    no slice-gradient zero/copy/add passes in backward.  h is either
+    already packed [2,*,d] (the fused edge-MLP kernel's layout — no
+    reshape in either direction) or row-major [*,2d] (fallback paths),
+    which the view+transpose covers (its backward pays one re-pack)."""
+    hb = h if h.dim() == 3 else h.view(-1, 2, d).transpose(0, 1)
     return torch.baddbmm(b2, hb, w2)
 
 

@@ -98,6 +98,30 @@ __global__ void k_gather_add3_v4(const float4* __restrict__ zs,
     }
 }
 
+// dz = (go_z ? go_z : 0) + go_h * silu'(z), where go_h is PACKED
+// [2, E, half] (plane-per-column-half, the fused edge-MLP h layout) and
+// z / dz are row-major [E, 2*half]
+__global__ void k_silu_bwd_packed(const float* __restrict__ go_p,
+                                  const float* __restrict__ go_z,
+                                  const float* __restrict__ z,
+                                  float* __restrict__ dz,
+                                  int64_t E, int32_t half) {
+    const int64_t total = E * 2 * half;
+    const int64_t plane = E * (int64_t)half;
+    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         t < total; t += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t e = t / (2 * half);
+        const int32_t j = (int32_t)(t - e * 2 * half);
+        const float gh = go_p[(j >= half ? plane : 0) + e * half
+                              + (j >= half ? j - half : j)];
+        const float zv = z[t];
+        const float sg = 1.0f / (1.0f + expf(-zv));
+        float g = gh * (sg * (1.0f + zv * (1.0f - sg)));
+        if (go_z) g += go_z[t];
+        dz[t] = g;
+    }
+}
+
 // dz = (go_z ? go_z : 0) + go_h * silu'(z) — fused SiLU backward
 __global__ void k_silu_bwd(const float* __restrict__ go_h,
                            const float* __restrict__ go_z,
@@ -144,84 +168,73 @@ __global__ void k_gather_add4_v4(const float4* __restrict__ z1,
 // Fused first-layer edge MLP (the gated-MLP split's per-edge GEMM folded
 // into the gather-add): for Din=64 -> Dout=128,
 //   z[e,:] = e_row[e,:] @ WT + bias + zs[src[e],:] + zd[dst[e],:]
-//   h      = silu(z)
-// WT ([64,128] row-major = linear.weight.T) + bias live in LDS for the
-// whole launch (32.5 KiB, loaded once per workgroup); each thread owns one
-// edge with its 64-float input row held in registers (the wave's loads
-// cover contiguous 16 KiB), outputs computed in 4 chunks of 32 with
-// LDS-broadcast weight reads.  Saves materializing the [E,128] GEMM
-// output: one full write + one full read of HBM per MLP vs rocBLAS + the
-// unfused gather-add (~5.2 GB/pass at li100k).
-// gzs/gzd/de backwards reuse the existing seg-sum / GEMM paths.
+//   h      = silu(z), emitted PACKED as [2, E, 64] (column-half planes)
+// Layout: one WAVE per edge, one LANE per pair of output columns
+// (l, l+64).  Each lane holds its two weight columns in 128 VGPRs (loaded
+// once per wave, L2-served); the edge index is wave-uniform
+// (readfirstlane) so the 64-float e-row comes through the SCALAR cache
+// and feeds v_fmac as the SGPR operand.  Every global access is
+// lane-contiguous: weight preload, zs/zd/zv row gathers (512 B/row), and
+// the z/h stores.  No LDS.  Saves materializing the [E,128] GEMM output
+// (~5.2 GB of HBM round-trip per MLP at li100k).
+// (Measured alternatives, run 21/22: LDS-broadcast weights were
+// issue-bound at 8.0 ms; two-edges-per-wave overflowed the SGPR file and
+// serialized at 4.1 ms; a one-column-per-lane split made the compiler
+// de-register the weight array.  This shape: 3.3 ms at li100k.)
 // ---------------------------------------------------------------------------
 template <int NGATHER>
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(256, 3)
 void k_edge_mlp_64x128(const float* __restrict__ erow,
-                       const float* __restrict__ WT,
-                       const float* __restrict__ bias,
-                       const float4* __restrict__ g0,   // [*,32] float4 rows
-                       const float4* __restrict__ g1,
-                       const float4* __restrict__ g2,   // NGATHER==3 only
+                       const float* __restrict__ WT,   // [64,128] row-major
+                       const float* __restrict__ bias, // [128]
+                       const float* __restrict__ g0,   // [*,128] rows
+                       const float* __restrict__ g1,
+                       const float* __restrict__ g2,   // NGATHER==3 only
                        const int32_t* __restrict__ i0,
                        const int32_t* __restrict__ i1,
                        const int32_t* __restrict__ i2,
-                       float4* __restrict__ out,
-                       float4* __restrict__ out_act,
+                       float* __restrict__ out,
+                       float* __restrict__ out_act,
                        int64_t E) {
-    __shared__ float w[64 * 128];
-    __shared__ float b[128];
-    for (int i = threadIdx.x; i < 64 * 128 / 4; i += blockDim.x)
-        ((float4*)w)[i] = ((const float4*)WT)[i];
-    for (int i = threadIdx.x; i < 128; i += blockDim.x) b[i] = bias[i];
-    __syncthreads();
-    for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-         idx < E; idx += (int64_t)gridDim.x * blockDim.x) {
-        float4 er[16];
-        const float4* ep = (const float4*)(erow + idx * 64);
+    const int lane = threadIdx.x & 63;
+    const int wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const int waves_per_block = blockDim.x >> 6;
+    float wlo[64], whi[64];
 #pragma unroll
-        for (int i = 0; i < 16; ++i) er[i] = ep[i];
-        const int64_t b0 = (int64_t)i0[idx] * 32;
-        const int64_t b1 = (int64_t)i1[idx] * 32;
-        const int64_t b2 = NGATHER == 3 ? (int64_t)i2[idx] * 32 : 0;
-#pragma clang loop unroll(disable)
-        for (int oc = 0; oc < 4; ++oc) {
-            float acc[32];
+    for (int k = 0; k < 64; ++k) {
+        wlo[k] = WT[k * 128 + lane];
+        whi[k] = WT[k * 128 + 64 + lane];
+    }
+    const float blo = bias[lane], bhi = bias[64 + lane];
+    for (int64_t edge = blockIdx.x * (int64_t)waves_per_block + wave;
+         edge < E; edge += (int64_t)gridDim.x * waves_per_block) {
+        const float* er = erow + edge * 64;    // wave-uniform -> scalar loads
+        float a0 = blo, a1 = bhi;
 #pragma unroll
-            for (int o = 0; o < 32; ++o) acc[o] = 0.0f;
-            // k fully unrolled via the float4 registers (dynamic indexing
-            // into er[] would spill it to scratch); oc stays a real loop
-            // to bound code size (~2k FMAs, not 8k)
-#pragma unroll
-            for (int i = 0; i < 16; ++i) {
-                const float4 ev = er[i];
-                const float* wr = w + (i * 4) * 128 + oc * 32;
-#pragma unroll
-                for (int o = 0; o < 32; ++o) {
-                    float t = fmaf(ev.x, wr[o], acc[o]);
-                    t = fmaf(ev.y, wr[128 + o], t);
-                    t = fmaf(ev.z, wr[256 + o], t);
-                    acc[o] = fmaf(ev.w, wr[384 + o], t);
-                }
-            }
-#pragma unroll
-            for (int o8 = 0; o8 < 8; ++o8) {
-                const int c4 = oc * 8 + o8;          // float4 col in [0,32)
-                const float4 a = g0[b0 + c4];
-                const float4 d = g1[b1 + c4];
-                float4 z4;
-                z4.x = acc[o8 * 4 + 0] + a.x + d.x + b[c4 * 4 + 0];
-                z4.y = acc[o8 * 4 + 1] + a.y + d.y + b[c4 * 4 + 1];
-                z4.z = acc[o8 * 4 + 2] + a.z + d.z + b[c4 * 4 + 2];
-                z4.w = acc[o8 * 4 + 3] + a.w + d.w + b[c4 * 4 + 3];
-                if (NGATHER == 3) {
-                    const float4 v = g2[b2 + c4];
-                    z4.x += v.x; z4.y += v.y; z4.z += v.z; z4.w += v.w;
-                }
-                out[idx * 32 + c4] = z4;
-                out_act[idx * 32 + c4] = make_float4(
-                    siluf(z4.x), siluf(z4.y), siluf(z4.z), siluf(z4.w));
-            }
+        for (int k = 0; k < 64; ++k) {
+            const float ek = er[k];
+            a0 = fmaf(ek, wlo[k], a0);
+            a1 = fmaf(ek, whi[k], a1);
         }
+        const float* r0 = g0 + (int64_t)i0[edge] * 128;
+        const float* r1 = g1 + (int64_t)i1[edge] * 128;
+        a0 += r0[lane] + r1[lane];
+        a1 += r0[64 + lane] + r1[64 + lane];
+        if (NGATHER == 3) {
+            const float* r2 = g2 + (int64_t)i2[edge] * 128;
+            a0 += r2[lane];
+            a1 += r2[64 + lane];
+        }
+        if (out) {                       // z saved for backward (row-major)
+            float* zp = out + edge * 128;
+            zp[lane] = a0;
+            zp[64 + lane] = a1;
+        }
+        // h is written PACKED [2, E, 64] (plane per column half): exactly
+        // the layout the second-layer batched GEMM consumes, so neither
+        // direction ever reshapes h (the [E,128] view would copy the grad)
+        out_act[edge * 64 + lane] = siluf(a0);
+        out_act[(E + edge) * 64 + lane] = siluf(a1);
     }
 }
 
@@ -651,9 +664,9 @@ int dm_edge_mlp3_f32(const float* erow, const float* WT, const float* bias,
         g_err = "edge_mlp fused kernel is compiled for Din=64, Dout=128";
         return -1;
     }
-    k_edge_mlp_64x128<2><<<nblocks(E, BLOCK), BLOCK, 0, s>>>(
-        erow, WT, bias, (const float4*)zs, (const float4*)zd, nullptr,
-        src, dst, nullptr, (float4*)out, (float4*)out_act, E);
+    // one wave per edge: BLOCK threads cover BLOCK/64 edges per iter
+    k_edge_mlp_64x128<2><<<nblocks(E, BLOCK / 64), BLOCK, 0, s>>>(
+        erow, WT, bias, zs, zd, nullptr, src, dst, nullptr, out, out_act, E);
     DM_CHECK_LAUNCH();
     return 0;
 }
@@ -668,10 +681,8 @@ int dm_edge_mlp4_f32(const float* arow, const float* WT, const float* bias,
         g_err = "edge_mlp fused kernel is compiled for Din=64, Dout=128";
         return -1;
     }
-    k_edge_mlp_64x128<3><<<nblocks(L, BLOCK), BLOCK, 0, s>>>(
-        arow, WT, bias, (const float4*)z1, (const float4*)z2,
-        (const float4*)zv, lsrc, ldst, center, (float4*)out, (float4*)out_act,
-        L);
+    k_edge_mlp_64x128<3><<<nblocks(L, BLOCK / 64), BLOCK, 0, s>>>(
+        arow, WT, bias, z1, z2, zv, lsrc, ldst, center, out, out_act, L);
     DM_CHECK_LAUNCH();
     return 0;
 }
@@ -812,6 +823,16 @@ int dm_rbf_env_bwd_f32(const float* go_exp, const float* d, const float* freqs,
     hipStream_t s = (hipStream_t)stream;
     k_rbf_env_bwd<<<nblocks(M, BLOCK), BLOCK, 0, s>>>(
         go_exp, d, freqs, cutoff, pexp, nrbf, gd, M);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_silu_bwd_packed_f32(const float* go_p, const float* go_z,
+                           const float* z, float* dz, int64_t E,
+                           int64_t half, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    k_silu_bwd_packed<<<nblocks(E * 2 * half, BLOCK), BLOCK, 0, s>>>(
+        go_p, go_z, z, dz, E, (int32_t)half);
     DM_CHECK_LAUNCH();
     return 0;
 }

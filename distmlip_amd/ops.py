@@ -45,6 +45,9 @@ def hip_lib() -> ctypes.CDLL:
                                          c_uint64]
         lib.dm_silu_bwd_f32.restype = c_int32
         lib.dm_silu_bwd_f32.argtypes = [fp, fp, fp, fp, c_int64, c_uint64]
+        lib.dm_silu_bwd_packed_f32.restype = c_int32
+        lib.dm_silu_bwd_packed_f32.argtypes = [fp, fp, fp, fp, c_int64,
+                                               c_int64, c_uint64]
         lib.dm_seg_sum_f32.restype = c_int32
         lib.dm_seg_sum_f32.argtypes = [fp, ip, fp, fp, c_int64, c_int64,
                                        c_uint64]
@@ -162,6 +165,16 @@ def raw_silu_bwd(go_h, go_z, z):
     return dz
 
 
+def raw_silu_bwd_packed(go_p, go_z, z):
+    """Same, with go_p in the fused edge-MLP's packed [2,E,half] h layout
+    (z and the returned dz are row-major [E,2*half])."""
+    dz = torch.empty_like(z)
+    _check(hip_lib().dm_silu_bwd_packed_f32(
+        _fp(go_p), _fp(go_z) if go_z is not None else None, _fp(z), _fp(dz),
+        z.shape[0], z.shape[1] // 2, _stream()), "dm_silu_bwd_packed_f32")
+    return dz
+
+
 class _GatherAdd3(torch.autograd.Function):
     """Emits (z, silu(z)) in one kernel; z doubles as the saved activation
     for the fused silu backward."""
@@ -220,13 +233,14 @@ class _EdgeMlp3(torch.autograd.Function):
         ctx.set_materialize_grads(False)
         ctx.n_nodes = zs.shape[0]
         E, dout = erow.shape[0], wt.shape[1]
-        z = torch.empty(E, dout, dtype=erow.dtype, device=erow.device)
-        needs_h = erow.requires_grad or zs.requires_grad or zd.requires_grad
-        h = torch.empty_like(z) if needs_h else z
+        h = torch.empty(2, E, dout // 2, dtype=erow.dtype, device=erow.device)
+        needs_z = erow.requires_grad or zs.requires_grad or zd.requires_grad
+        z = (torch.empty(E, dout, dtype=erow.dtype, device=erow.device)
+             if needs_z else h[0, :0])    # no backward -> skip the z write
         _check(hip_lib().dm_edge_mlp3_f32(
             _fp(erow), _fp(wt), _fp(bias), _fp(zs), _fp(zd), _ip(src),
-            _ip(dst), _fp(z), _fp(h), E, wt.shape[0], dout, _stream()),
-            "dm_edge_mlp3_f32")
+            _ip(dst), _fp(z) if needs_z else None, _fp(h), E, wt.shape[0],
+            dout, _stream()), "dm_edge_mlp3_f32")
         ctx.save_for_backward(src_perm, src_row_ptr, row_ptr, z, wt)
         return z, h
 
@@ -234,9 +248,9 @@ class _EdgeMlp3(torch.autograd.Function):
     def backward(ctx, go_z, go_h):
         src_perm, src_row_ptr, row_ptr, z, wt = ctx.saved_tensors
         if go_h is not None:
-            dz = raw_silu_bwd(go_h.contiguous(),
-                              go_z.contiguous() if go_z is not None else None,
-                              z)
+            dz = raw_silu_bwd_packed(
+                go_h.contiguous(),
+                go_z.contiguous() if go_z is not None else None, z)
         else:
             dz = go_z.contiguous()
         de = dz @ wt.t()
@@ -258,14 +272,16 @@ class _EdgeMlp4(torch.autograd.Function):
         ctx.n_bonds = z1.shape[0]
         ctx.n_nodes = zv.shape[0]
         L, dout = arow.shape[0], wt.shape[1]
-        z = torch.empty(L, dout, dtype=arow.dtype, device=arow.device)
-        needs_h = (arow.requires_grad or z1.requires_grad or z2.requires_grad
+        h = torch.empty(2, L, dout // 2, dtype=arow.dtype, device=arow.device)
+        needs_z = (arow.requires_grad or z1.requires_grad or z2.requires_grad
                    or zv.requires_grad)
-        h = torch.empty_like(z) if needs_h else z
+        z = (torch.empty(L, dout, dtype=arow.dtype, device=arow.device)
+             if needs_z else h[0, :0])
         _check(hip_lib().dm_edge_mlp4_f32(
             _fp(arow), _fp(wt), _fp(bias), _fp(z1), _fp(z2), _fp(zv),
-            _ip(pd.l_src), _ip(pd.l_dst), _ip(pd.center), _fp(z), _fp(h), L,
-            wt.shape[0], dout, _stream()), "dm_edge_mlp4_f32")
+            _ip(pd.l_src), _ip(pd.l_dst), _ip(pd.center),
+            _fp(z) if needs_z else None, _fp(h), L, wt.shape[0], dout,
+            _stream()), "dm_edge_mlp4_f32")
         ctx.save_for_backward(z, wt)
         return z, h
 
@@ -274,9 +290,9 @@ class _EdgeMlp4(torch.autograd.Function):
         pd = ctx.pd
         z, wt = ctx.saved_tensors
         if go_h is not None:
-            dz = raw_silu_bwd(go_h.contiguous(),
-                              go_z.contiguous() if go_z is not None else None,
-                              z)
+            dz = raw_silu_bwd_packed(
+                go_h.contiguous(),
+                go_z.contiguous() if go_z is not None else None, z)
         else:
             dz = go_z.contiguous()
         da = dz @ wt.t()
@@ -508,7 +524,8 @@ class HipOps:
 
     def edge_mlp3_act(self, erow, wt, bias, zs, zd, pd):
         """silu(erow @ wt + bias + zs[src] + zd[dst]) fused in one kernel
-        (wt = first-layer weight.T, [64,128] only)."""
+        (wt = first-layer weight.T, [64,128] only).  Returns h PACKED
+        [2, E, 64] — feed it straight to the second-layer batched GEMM."""
         _z, h = _EdgeMlp3.apply(erow.contiguous(), wt, bias, zs.contiguous(),
                                 zd.contiguous(), pd.src, pd.dst, pd.src_perm,
                                 pd.src_row_ptr, pd.row_ptr)
