@@ -98,30 +98,6 @@ __global__ void k_gather_add3_v4(const float4* __restrict__ zs,
     }
 }
 
-// dz = (go_z ? go_z : 0) + go_h * silu'(z), where go_h is PACKED
-// [2, E, half] (plane-per-column-half, the fused edge-MLP h layout) and
-// z / dz are row-major [E, 2*half]
-__global__ void k_silu_bwd_packed(const float* __restrict__ go_p,
-                                  const float* __restrict__ go_z,
-                                  const float* __restrict__ z,
-                                  float* __restrict__ dz,
-                                  int64_t E, int32_t half) {
-    const int64_t total = E * 2 * half;
-    const int64_t plane = E * (int64_t)half;
-    for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-         t < total; t += (int64_t)gridDim.x * blockDim.x) {
-        const int64_t e = t / (2 * half);
-        const int32_t j = (int32_t)(t - e * 2 * half);
-        const float gh = go_p[(j >= half ? plane : 0) + e * half
-                              + (j >= half ? j - half : j)];
-        const float zv = z[t];
-        const float sg = 1.0f / (1.0f + expf(-zv));
-        float g = gh * (sg * (1.0f + zv * (1.0f - sg)));
-        if (go_z) g += go_z[t];
-        dz[t] = g;
-    }
-}
-
 // dz = (go_z ? go_z : 0) + go_h * silu'(z) — fused SiLU backward
 __global__ void k_silu_bwd(const float* __restrict__ go_h,
                            const float* __restrict__ go_z,
@@ -168,7 +144,7 @@ __global__ void k_gather_add4_v4(const float4* __restrict__ z1,
 // Fused first-layer edge MLP (the gated-MLP split's per-edge GEMM folded
 // into the gather-add): for Din=64 -> Dout=128,
 //   z[e,:] = e_row[e,:] @ WT + bias + zs[src[e],:] + zd[dst[e],:]
-//   h      = silu(z), emitted PACKED as [2, E, 64] (column-half planes)
+//   h      = silu(z)
 // Layout: one WAVE per edge, one LANE per pair of output columns
 // (l, l+64).  Each lane holds its two weight columns in 128 VGPRs (loaded
 // once per wave, L2-served); the edge index is wave-uniform
@@ -225,16 +201,17 @@ void k_edge_mlp_64x128(const float* __restrict__ erow,
             a0 += r2[lane];
             a1 += r2[64 + lane];
         }
-        if (out) {                       // z saved for backward (row-major)
-            float* zp = out + edge * 128;
+        if (out) {                       // z saved for backward; skipped
+            float* zp = out + edge * 128;  // in no-grad passes
             zp[lane] = a0;
             zp[64 + lane] = a1;
         }
-        // h is written PACKED [2, E, 64] (plane per column half): exactly
-        // the layout the second-layer batched GEMM consumes, so neither
-        // direction ever reshapes h (the [E,128] view would copy the grad)
-        out_act[edge * 64 + lane] = siluf(a0);
-        out_act[(E + edge) * 64 + lane] = siluf(a1);
+        // (a packed [2,E,64] h layout was tried to feed the second-layer
+        // bmm without a reshape, but the two distant 256 B plane stores
+        // cost more than the saved copies — profiles r21 vs r23)
+        float* hp = out_act + edge * 128;
+        hp[lane] = siluf(a0);
+        hp[64 + lane] = siluf(a1);
     }
 }
 
@@ -823,16 +800,6 @@ int dm_rbf_env_bwd_f32(const float* go_exp, const float* d, const float* freqs,
     hipStream_t s = (hipStream_t)stream;
     k_rbf_env_bwd<<<nblocks(M, BLOCK), BLOCK, 0, s>>>(
         go_exp, d, freqs, cutoff, pexp, nrbf, gd, M);
-    DM_CHECK_LAUNCH();
-    return 0;
-}
-
-int dm_silu_bwd_packed_f32(const float* go_p, const float* go_z,
-                           const float* z, float* dz, int64_t E,
-                           int64_t half, uint64_t stream) {
-    hipStream_t s = (hipStream_t)stream;
-    k_silu_bwd_packed<<<nblocks(E * 2 * half, BLOCK), BLOCK, 0, s>>>(
-        go_p, go_z, z, dz, E, (int32_t)half);
     DM_CHECK_LAUNCH();
     return 0;
 }

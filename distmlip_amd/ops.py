@@ -45,9 +45,6 @@ def hip_lib() -> ctypes.CDLL:
                                          c_uint64]
         lib.dm_silu_bwd_f32.restype = c_int32
         lib.dm_silu_bwd_f32.argtypes = [fp, fp, fp, fp, c_int64, c_uint64]
-        lib.dm_silu_bwd_packed_f32.restype = c_int32
-        lib.dm_silu_bwd_packed_f32.argtypes = [fp, fp, fp, fp, c_int64,
-                                               c_int64, c_uint64]
         lib.dm_seg_sum_f32.restype = c_int32
         lib.dm_seg_sum_f32.argtypes = [fp, ip, fp, fp, c_int64, c_int64,
                                        c_uint64]
@@ -165,6 +162,33 @@ def raw_silu_bwd(go_h, go_z, z):
     return dz
 
 
+class _Gather(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, idx, perm, row_ptr):
+        ctx.perm = perm          # None when idx is the sorted direction
+        ctx.row_ptr = row_ptr
+        ctx.n_rows = x.shape[0]
+        return raw_gather(x, idx)
+
+    @staticmethod
+    def backward(ctx, grad):
+        grad = grad.contiguous()
+        if ctx.perm is None:
+            gx = raw_seg_sum(grad, ctx.row_ptr, ctx.n_rows)
+        else:
+            gx = raw_seg_sum_gather(grad, ctx.perm, ctx.row_ptr, ctx.n_rows)
+        return gx, None, None, None
+
+
+def raw_silu_bwd(go_h, go_z, z):
+    """dz = (go_z or 0) + go_h * silu'(z), one fused pass."""
+    dz = torch.empty_like(z)
+    _check(hip_lib().dm_silu_bwd_f32(
+        _fp(go_h), _fp(go_z) if go_z is not None else None, _fp(z), _fp(dz),
+        z.numel(), _stream()), "dm_silu_bwd_f32")
+    return dz
+
+
 def raw_silu_bwd_packed(go_p, go_z, z):
     """Same, with go_p in the fused edge-MLP's packed [2,E,half] h layout
     (z and the returned dz are row-major [E,2*half])."""
@@ -233,10 +257,9 @@ class _EdgeMlp3(torch.autograd.Function):
         ctx.set_materialize_grads(False)
         ctx.n_nodes = zs.shape[0]
         E, dout = erow.shape[0], wt.shape[1]
-        h = torch.empty(2, E, dout // 2, dtype=erow.dtype, device=erow.device)
+        h = torch.empty(E, dout, dtype=erow.dtype, device=erow.device)
         needs_z = erow.requires_grad or zs.requires_grad or zd.requires_grad
-        z = (torch.empty(E, dout, dtype=erow.dtype, device=erow.device)
-             if needs_z else h[0, :0])    # no backward -> skip the z write
+        z = torch.empty_like(h) if needs_z else h[:0]  # no bwd: skip z write
         _check(hip_lib().dm_edge_mlp3_f32(
             _fp(erow), _fp(wt), _fp(bias), _fp(zs), _fp(zd), _ip(src),
             _ip(dst), _fp(z) if needs_z else None, _fp(h), E, wt.shape[0],
@@ -248,9 +271,9 @@ class _EdgeMlp3(torch.autograd.Function):
     def backward(ctx, go_z, go_h):
         src_perm, src_row_ptr, row_ptr, z, wt = ctx.saved_tensors
         if go_h is not None:
-            dz = raw_silu_bwd_packed(
-                go_h.contiguous(),
-                go_z.contiguous() if go_z is not None else None, z)
+            dz = raw_silu_bwd(go_h.contiguous(),
+                              go_z.contiguous() if go_z is not None else None,
+                              z)
         else:
             dz = go_z.contiguous()
         de = dz @ wt.t()
@@ -272,11 +295,10 @@ class _EdgeMlp4(torch.autograd.Function):
         ctx.n_bonds = z1.shape[0]
         ctx.n_nodes = zv.shape[0]
         L, dout = arow.shape[0], wt.shape[1]
-        h = torch.empty(2, L, dout // 2, dtype=arow.dtype, device=arow.device)
+        h = torch.empty(L, dout, dtype=arow.dtype, device=arow.device)
         needs_z = (arow.requires_grad or z1.requires_grad or z2.requires_grad
                    or zv.requires_grad)
-        z = (torch.empty(L, dout, dtype=arow.dtype, device=arow.device)
-             if needs_z else h[0, :0])
+        z = torch.empty_like(h) if needs_z else h[:0]
         _check(hip_lib().dm_edge_mlp4_f32(
             _fp(arow), _fp(wt), _fp(bias), _fp(z1), _fp(z2), _fp(zv),
             _ip(pd.l_src), _ip(pd.l_dst), _ip(pd.center),
@@ -290,9 +312,9 @@ class _EdgeMlp4(torch.autograd.Function):
         pd = ctx.pd
         z, wt = ctx.saved_tensors
         if go_h is not None:
-            dz = raw_silu_bwd_packed(
-                go_h.contiguous(),
-                go_z.contiguous() if go_z is not None else None, z)
+            dz = raw_silu_bwd(go_h.contiguous(),
+                              go_z.contiguous() if go_z is not None else None,
+                              z)
         else:
             dz = go_z.contiguous()
         da = dz @ wt.t()
@@ -524,8 +546,7 @@ class HipOps:
 
     def edge_mlp3_act(self, erow, wt, bias, zs, zd, pd):
         """silu(erow @ wt + bias + zs[src] + zd[dst]) fused in one kernel
-        (wt = first-layer weight.T, [64,128] only).  Returns h PACKED
-        [2, E, 64] — feed it straight to the second-layer batched GEMM."""
+        (wt = first-layer weight.T, [64,128] only)."""
         _z, h = _EdgeMlp3.apply(erow.contiguous(), wt, bias, zs.contiguous(),
                                 zd.contiguous(), pd.src, pd.dst, pd.src_perm,
                                 pd.src_row_ptr, pd.row_ptr)

@@ -50,12 +50,6 @@ int dm_gather_add4_f32(const float* z1, const float* z2, const float* za,
                        float* out_act, int64_t L, int64_t D, uint64_t stream);
 
 /* dz = (go_z ? go_z : 0) + go_h * silu'(z) — backward of the fused silu */
-/* packed-gradient SiLU backward: go_p is [2, E, half] (the fused
- * edge-MLP's h layout); z and dz are row-major [E, 2*half]. */
-int dm_silu_bwd_packed_f32(const float* go_p, const float* go_z,
-                           const float* z, float* dz, int64_t E,
-                           int64_t half, uint64_t stream);
-
 int dm_silu_bwd_f32(const float* go_h, const float* go_z, const float* z,
                     float* dz, int64_t total, uint64_t stream);
 
@@ -81,9 +75,7 @@ int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
  * other shapes return an error and the binding falls back to the unfused
  * GEMM + dm_gather_add{3,4} path.  out (= z, row-major [E,128], the
  * backward save) may be NULL to skip the write in no-grad passes;
- * out_act (= silu(z)) is written PACKED [2, E, 64] — column-half planes,
- * the layout the second-layer batched GEMM consumes, see
- * dm_silu_bwd_packed_f32 for its gradient. */
+ * out_act = silu(z), row-major. */
 int dm_edge_mlp3_f32(const float* erow, const float* WT, const float* bias,
                      const float* zs, const float* zd, const int32_t* src,
                      const int32_t* dst, float* out, float* out_act,

@@ -237,14 +237,12 @@ def test_edge_mlp_fused_vs_unfused():
     zs2 = zs.detach().clone().requires_grad_(True)
     zd2 = zd.detach().clone().requires_grad_(True)
     zref = e2 @ wt + bias + zs2[src64] + zd2[dst64]
-    # h is PACKED [2, E, Dout/2] (column-half planes)
-    hrefp = torch.stack(torch.nn.functional.silu(zref).chunk(2, dim=1))
+    href = torch.nn.functional.silu(zref)
     assert torch.allclose(z, zref, atol=2e-4), (z - zref).abs().max()
-    assert h.shape == (2, E, Dout // 2)
-    assert torch.allclose(h, hrefp, atol=2e-4)
+    assert torch.allclose(h, href, atol=2e-4)
     go = torch.randn_like(h)
     h.backward(go)
-    hrefp.backward(go)
+    href.backward(go)
     assert torch.allclose(erow.grad, e2.grad, atol=2e-3), \
         (erow.grad - e2.grad).abs().max()
     assert torch.allclose(zs.grad, zs2.grad, atol=2e-3)
@@ -271,8 +269,8 @@ def test_edge_mlp_fused_vs_unfused():
     _z4, h4 = _EdgeMlp4.apply(a, wt, bias, z1, z2, zv, pd)
     a2 = a.detach().clone().requires_grad_(True)
     z1b = z1.detach().clone().requires_grad_(True)
-    href4 = torch.stack(torch.nn.functional.silu(
-        a2 @ wt + bias + z1b[l_src64] + z2[l_dst64] + zv[c64]).chunk(2, dim=1))
+    href4 = torch.nn.functional.silu(
+        a2 @ wt + bias + z1b[l_src64] + z2[l_dst64] + zv[c64])
     assert torch.allclose(h4, href4, atol=2e-4)
     go4 = torch.randn_like(h4)
     h4.backward(go4)
