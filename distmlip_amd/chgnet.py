@@ -24,6 +24,7 @@ fp summation order.
 """
 from __future__ import annotations
 
+import os
 from copy import deepcopy
 from typing import List, Optional
 
@@ -94,13 +95,28 @@ def _second_layer_packed(h, w2, b2, d: int):
     return torch.baddbmm(b2, hb, w2)
 
 
+def _use_fused() -> bool:
+    """Fused first-layer kernel policy.  Measured on li100k (same box,
+    run 25): in the GRAD-RECORDING path the fused kernel loses ~8%
+    step time to rocBLAS+gather-add (172 vs 187 ms), but in NO-GRAD
+    passes (checkpoint outer forward, inference) it skips the z save and
+    the [E,128] GEMM round-trip entirely (~4 GB vs ~9 GB of HBM per MLP),
+    so: fused iff grad is off.  DM_FUSED_MLP=1 / DM_NO_FUSED_MLP=1
+    force it on / off for A/B runs."""
+    if os.environ.get("DM_NO_FUSED_MLP", "0") == "1":
+        return False
+    if os.environ.get("DM_FUSED_MLP", "0") == "1":
+        return True
+    return not torch.is_grad_enabled()
+
+
 def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
     """GatedMLP over cat(v[src], v[dst], e) via split-linear + one 2h-wide
     gather_add3, finished by the fused gated-combine epilogue:
     returns base + silu(core2(silu(z_c))) * sigmoid(gate2(silu(z_g))) * w."""
     wcg, bcg, w2, b2, fusedT = _packed_weights(mlp)
     ws, wd = wcg[:, :d], wcg[:, d:2 * d]
-    if (d == 64 and not wcg.requires_grad
+    if (d == 64 and not wcg.requires_grad and _use_fused()
             and hasattr(ops, "edge_mlp3_act")):
         # per-edge GEMM fused into the gather kernel (no [E,2d] transient)
         h = ops.edge_mlp3_act(e, fusedT, bcg, v @ ws.t(), v @ wd.t(), pd)
@@ -116,7 +132,7 @@ def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
     """GatedMLP over cat(n[l_src], n[l_dst], a, v[center]), same structure."""
     wcg, bcg, w2p, b2p, fusedT = _packed_weights(mlp)
     w1, w2, wv = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 3 * d:]
-    if (d == 64 and not wcg.requires_grad
+    if (d == 64 and not wcg.requires_grad and _use_fused()
             and hasattr(ops, "edge_mlp4_act")):
         h = ops.edge_mlp4_act(a, fusedT, bcg, n @ w1.t(), n @ w2.t(),
                               v @ wv.t(), pd)
