@@ -95,19 +95,25 @@ def _second_layer_packed(h, w2, b2, d: int):
     return torch.baddbmm(b2, hb, w2)
 
 
-def _use_fused() -> bool:
+_FUSED_BIG_ROWS = 40_000_000
+
+
+def _use_fused(rows: int) -> bool:
     """Fused first-layer kernel policy.  Measured on li100k (same box,
     run 25): in the GRAD-RECORDING path the fused kernel loses ~8%
     step time to rocBLAS+gather-add (172 vs 187 ms), but in NO-GRAD
     passes (checkpoint outer forward, inference) it skips the z save and
     the [E,128] GEMM round-trip entirely (~4 GB vs ~9 GB of HBM per MLP),
-    so: fused iff grad is off.  DM_FUSED_MLP=1 / DM_NO_FUSED_MLP=1
-    force it on / off for A/B runs."""
+    so: fused iff grad is off — plus, for HUGE graphs (rows > 40M, where
+    a single [E,128] fp32 tensor is >20 GB), fused in the recording path
+    too: it materializes one fewer such transient per MLP, which is what
+    decides whether 2M atoms fit in 288 GB.  DM_FUSED_MLP=1 /
+    DM_NO_FUSED_MLP=1 force it on / off for A/B runs."""
     if os.environ.get("DM_NO_FUSED_MLP", "0") == "1":
         return False
     if os.environ.get("DM_FUSED_MLP", "0") == "1":
         return True
-    return not torch.is_grad_enabled()
+    return not torch.is_grad_enabled() or rows > _FUSED_BIG_ROWS
 
 
 def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
@@ -116,7 +122,7 @@ def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
     returns base + silu(core2(silu(z_c))) * sigmoid(gate2(silu(z_g))) * w."""
     wcg, bcg, w2, b2, fusedT = _packed_weights(mlp)
     ws, wd = wcg[:, :d], wcg[:, d:2 * d]
-    if (d == 64 and not wcg.requires_grad and _use_fused()
+    if (d == 64 and not wcg.requires_grad and _use_fused(e.shape[0])
             and hasattr(ops, "edge_mlp3_act")):
         # per-edge GEMM fused into the gather kernel (no [E,2d] transient)
         h = ops.edge_mlp3_act(e, fusedT, bcg, v @ ws.t(), v @ wd.t(), pd)
@@ -132,7 +138,7 @@ def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
     """GatedMLP over cat(n[l_src], n[l_dst], a, v[center]), same structure."""
     wcg, bcg, w2p, b2p, fusedT = _packed_weights(mlp)
     w1, w2, wv = wcg[:, :d], wcg[:, d:2 * d], wcg[:, 3 * d:]
-    if (d == 64 and not wcg.requires_grad and _use_fused()
+    if (d == 64 and not wcg.requires_grad and _use_fused(a.shape[0])
             and hasattr(ops, "edge_mlp4_act")):
         h = ops.edge_mlp4_act(a, fusedT, bcg, n @ w1.t(), n @ w2.t(),
                               v @ wv.t(), pd)

@@ -343,6 +343,35 @@ def test_e2e_energy_forces_vs_oracle(P):
 
 
 @requires_gpu
+def test_fused_policy_equivalence():
+    """Both first-layer paths (fused edge-MLP kernel vs rocBLAS +
+    gather_add) must agree end-to-end through the engine — guards the
+    grad-mode/size policy branch in chgnet._use_fused."""
+    import os
+
+    from distmlip_amd.model import CHGNetCore
+    from distmlip_amd.runtime import SpmdEngine
+    from distmlip_amd.structures import diamond_si
+
+    s = diamond_si((10, 2, 2), jitter=0.1, seed=5)
+    core = CHGNetCore.seeded(seed=0)
+    outs = {}
+    for mode in ("DM_FUSED_MLP", "DM_NO_FUSED_MLP"):
+        os.environ[mode] = "1"
+        try:
+            eng = SpmdEngine(core.float(), world=1, threads=4)
+            outs[mode] = eng.step(s)
+        finally:
+            os.environ.pop(mode, None)
+    dE = abs(outs["DM_FUSED_MLP"]["energy"].item()
+             - outs["DM_NO_FUSED_MLP"]["energy"].item())
+    dF = (outs["DM_FUSED_MLP"]["forces_owned"]
+          - outs["DM_NO_FUSED_MLP"]["forces_owned"]).abs().max().item()
+    assert dE < 1e-3, dE
+    assert dF < 2e-4, dF
+
+
+@requires_gpu
 def test_spmd_engine_world1_vs_oracle():
     """The bench/production path (SpmdEngine at world=1, per-rank geometry,
     HIP kernels) against the fp64 oracle."""
