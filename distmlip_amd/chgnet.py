@@ -85,13 +85,36 @@ def _packed_weights(mlp: GatedMLP):
     return packed
 
 
+class _SecondLayer(torch.autograd.Function):
+    """cg [2,*,d] = baddbmm over the packed hidden halves of h [*,2d].
+    Hand-written backward: two GEMMs writing STRIDED into one row-major
+    dh (ldc = 2d), so the [*,2d] re-pack copy autograd's view+transpose
+    backward would insert never happens.  w2/b2 are frozen packs
+    (_packed_weights) — no weight grads."""
+
+    @staticmethod
+    def forward(ctx, h, w2, b2):
+        d = w2.shape[1]
+        hb = h.view(-1, 2, d).transpose(0, 1)
+        ctx.save_for_backward(w2)
+        return torch.baddbmm(b2, hb, w2)
+
+    @staticmethod
+    def backward(ctx, dcg):
+        (w2,) = ctx.saved_tensors
+        d = w2.shape[1]
+        dcg = dcg.contiguous()
+        dh = torch.empty(dcg.shape[1], 2 * d, dtype=dcg.dtype,
+                         device=dcg.device)
+        torch.mm(dcg[0], w2[0].t(), out=dh[:, :d])
+        torch.mm(dcg[1], w2[1].t(), out=dh[:, d:])
+        return dh, None, None
+
+
 def _second_layer_packed(h, w2, b2, d: int):
-    """cg [2,*,d] = baddbmm over the packed hidden halves; no slicing, so
-    no slice-gradient zero/copy/add passes in backward.  h is either
-    already packed [2,*,d] (the fused edge-MLP kernel's layout — no
-    reshape in either direction) or row-major [*,2d] (fallback paths),
-    which the view+transpose covers (its backward pays one re-pack)."""
-    hb = h if h.dim() == 3 else h.view(-1, 2, d).transpose(0, 1)
+    if h.requires_grad and not w2.requires_grad:
+        return _SecondLayer.apply(h, w2, b2)
+    hb = h.view(-1, 2, d).transpose(0, 1)
     return torch.baddbmm(b2, hb, w2)
 
 
