@@ -151,7 +151,8 @@ def gated_mlp_split3(mlp: GatedMLP, v, e, pd, ops, d: int, w=None, base=None):
         h = ops.edge_mlp3_act(e, fusedT, bcg, v @ ws.t(), v @ wd.t(), pd)
     else:
         we = wcg[:, 2 * d:]
-        h = ops.gather_add3_act(v @ ws.t(), v @ wd.t(), e @ we.t() + bcg, pd)
+        h = ops.gather_add3_act(v @ ws.t(), v @ wd.t(),
+                                torch.addmm(bcg, e, we.t()), pd)
     cg = _second_layer_packed(h, w2, b2, d)
     return ops.gated_combine_packed(cg, w, base)
 
@@ -167,8 +168,8 @@ def gated_mlp_split4(mlp: GatedMLP, n, a, v, pd, ops, d: int, w=None,
                               v @ wv.t(), pd)
     else:
         wa = wcg[:, 2 * d:3 * d]
-        h = ops.gather_add4_act(n @ w1.t(), n @ w2.t(), a @ wa.t() + bcg,
-                                v @ wv.t(), pd)
+        h = ops.gather_add4_act(n @ w1.t(), n @ w2.t(),
+                                torch.addmm(bcg, a, wa.t()), v @ wv.t(), pd)
     cg = _second_layer_packed(h, w2p, b2p, d)
     return ops.gated_combine_packed(cg, w, base)
 
