@@ -221,31 +221,32 @@ void k_edge_mlp_64x128(const float* __restrict__ erow,
 // messages are CONTIGUOUS float4s, so each loop iteration is a fully
 // coalesced 256 B read per group (1 KiB per wave).
 // ---------------------------------------------------------------------------
+template <int LANE_BITS>
 __global__ void k_seg_sum_v4(const float4* __restrict__ msg,
                              const int32_t* __restrict__ rp,
                              const float4* __restrict__ base,
                              float4* __restrict__ out,
                              int64_t N, int32_t D4) {
-    const int lane16 = threadIdx.x & 15;
-    const int group = threadIdx.x >> 4;              // 16 groups per block
-    const int groups_per_block = blockDim.x >> 4;
+    const int lane = threadIdx.x & ((1 << LANE_BITS) - 1);
+    const int group = threadIdx.x >> LANE_BITS;
+    const int groups_per_block = blockDim.x >> LANE_BITS;
     for (int64_t row = blockIdx.x * (int64_t)groups_per_block + group;
          row < N; row += (int64_t)gridDim.x * groups_per_block) {
-        // D4 <= 16 assumed for this kernel (D == 64); lane c covers col c
-        if (lane16 < D4) {
+        // D4 <= (1<<LANE_BITS); lane c covers float4 column c
+        if (lane < D4) {
             // fp64 accumulate (register-only; kernel is HBM-bound so the
             // extra VALU is free) — tightens force parity vs the oracle
             double ax = 0, ay = 0, az = 0, aw = 0;
             if (base) {
-                const float4 b = base[row * D4 + lane16];
+                const float4 b = base[row * D4 + lane];
                 ax = b.x; ay = b.y; az = b.z; aw = b.w;
             }
             const int32_t lo = rp[row], hi = rp[row + 1];
             for (int32_t j = lo; j < hi; ++j) {
-                const float4 m = msg[(int64_t)j * D4 + lane16];
+                const float4 m = msg[(int64_t)j * D4 + lane];
                 ax += m.x; ay += m.y; az += m.z; aw += m.w;
             }
-            out[row * D4 + lane16] = make_float4((float)ax, (float)ay,
+            out[row * D4 + lane] = make_float4((float)ax, (float)ay,
                                                  (float)az, (float)aw);
         }
     }
@@ -253,39 +254,40 @@ __global__ void k_seg_sum_v4(const float4* __restrict__ msg,
 
 // unroll-4 variant of the D4<=16 segmented sum: keeps 4 independent row
 // reads in flight per group iteration (A/B candidate vs k_seg_sum_v4)
+template <int LANE_BITS>
 __global__ void k_seg_sum_v4_u4(const float4* __restrict__ msg,
                                 const int32_t* __restrict__ rp,
                                 const float4* __restrict__ base,
                                 float4* __restrict__ out,
                                 int64_t N, int32_t D4) {
-    const int lane16 = threadIdx.x & 15;
-    const int group = threadIdx.x >> 4;
-    const int groups_per_block = blockDim.x >> 4;
+    const int lane = threadIdx.x & ((1 << LANE_BITS) - 1);
+    const int group = threadIdx.x >> LANE_BITS;
+    const int groups_per_block = blockDim.x >> LANE_BITS;
     for (int64_t row = blockIdx.x * (int64_t)groups_per_block + group;
          row < N; row += (int64_t)gridDim.x * groups_per_block) {
-        if (lane16 < D4) {
+        if (lane < D4) {
             double ax = 0, ay = 0, az = 0, aw = 0;
             if (base) {
-                const float4 b = base[row * D4 + lane16];
+                const float4 b = base[row * D4 + lane];
                 ax = b.x; ay = b.y; az = b.z; aw = b.w;
             }
             const int32_t lo = rp[row], hi = rp[row + 1];
             int32_t j = lo;
             for (; j + 4 <= hi; j += 4) {
-                const float4 m0 = msg[(int64_t)(j + 0) * D4 + lane16];
-                const float4 m1 = msg[(int64_t)(j + 1) * D4 + lane16];
-                const float4 m2 = msg[(int64_t)(j + 2) * D4 + lane16];
-                const float4 m3 = msg[(int64_t)(j + 3) * D4 + lane16];
+                const float4 m0 = msg[(int64_t)(j + 0) * D4 + lane];
+                const float4 m1 = msg[(int64_t)(j + 1) * D4 + lane];
+                const float4 m2 = msg[(int64_t)(j + 2) * D4 + lane];
+                const float4 m3 = msg[(int64_t)(j + 3) * D4 + lane];
                 ax += m0.x; ay += m0.y; az += m0.z; aw += m0.w;
                 ax += m1.x; ay += m1.y; az += m1.z; aw += m1.w;
                 ax += m2.x; ay += m2.y; az += m2.z; aw += m2.w;
                 ax += m3.x; ay += m3.y; az += m3.z; aw += m3.w;
             }
             for (; j < hi; ++j) {
-                const float4 m = msg[(int64_t)j * D4 + lane16];
+                const float4 m = msg[(int64_t)j * D4 + lane];
                 ax += m.x; ay += m.y; az += m.z; aw += m.w;
             }
-            out[row * D4 + lane16] = make_float4((float)ax, (float)ay,
+            out[row * D4 + lane] = make_float4((float)ax, (float)ay,
                                                  (float)az, (float)aw);
         }
     }
@@ -310,29 +312,30 @@ __global__ void k_seg_sum_s(const float* __restrict__ msg,
     }
 }
 
+template <int LANE_BITS>
 __global__ void k_seg_sum_gather_v4(const float4* __restrict__ msg,
                                     const int32_t* __restrict__ perm,
                                     const int32_t* __restrict__ rp,
                                     const float4* __restrict__ base,
                                     float4* __restrict__ out,
                                     int64_t N, int32_t D4) {
-    const int lane16 = threadIdx.x & 15;
-    const int group = threadIdx.x >> 4;
-    const int groups_per_block = blockDim.x >> 4;
+    const int lane = threadIdx.x & ((1 << LANE_BITS) - 1);
+    const int group = threadIdx.x >> LANE_BITS;
+    const int groups_per_block = blockDim.x >> LANE_BITS;
     for (int64_t row = blockIdx.x * (int64_t)groups_per_block + group;
          row < N; row += (int64_t)gridDim.x * groups_per_block) {
-        if (lane16 < D4) {
+        if (lane < D4) {
             double ax = 0, ay = 0, az = 0, aw = 0;
             if (base) {
-                const float4 b = base[row * D4 + lane16];
+                const float4 b = base[row * D4 + lane];
                 ax = b.x; ay = b.y; az = b.z; aw = b.w;
             }
             const int32_t lo = rp[row], hi = rp[row + 1];
             for (int32_t j = lo; j < hi; ++j) {
-                const float4 m = msg[(int64_t)perm[j] * D4 + lane16];
+                const float4 m = msg[(int64_t)perm[j] * D4 + lane];
                 ax += m.x; ay += m.y; az += m.z; aw += m.w;
             }
-            out[row * D4 + lane16] = make_float4((float)ax, (float)ay,
+            out[row * D4 + lane] = make_float4((float)ax, (float)ay,
                                                  (float)az, (float)aw);
         }
     }
@@ -705,13 +708,17 @@ int dm_seg_sum_f32(const float* msg, const int32_t* row_ptr,
     hipStream_t s = (hipStream_t)stream;
     if (D % 4 == 0 && D / 4 <= 16) {
         if (g_seg_variant == 1)
-            k_seg_sum_v4_u4<<<nblocks(N, 16), BLOCK, 0, s>>>(
+            k_seg_sum_v4_u4<4><<<nblocks(N, 16), BLOCK, 0, s>>>(
                 (const float4*)msg, row_ptr, (const float4*)base,
                 (float4*)out, N, (int32_t)(D / 4));
         else
-            k_seg_sum_v4<<<nblocks(N, 16), BLOCK, 0, s>>>(
+            k_seg_sum_v4<4><<<nblocks(N, 16), BLOCK, 0, s>>>(
                 (const float4*)msg, row_ptr, (const float4*)base,
                 (float4*)out, N, (int32_t)(D / 4));
+    } else if (D % 4 == 0 && D / 4 <= 32) {      // D == 128: the backward
+        k_seg_sum_v4_u4<5><<<nblocks(N, 8), BLOCK, 0, s>>>(   // dz sums
+            (const float4*)msg, row_ptr, (const float4*)base,
+            (float4*)out, N, (int32_t)(D / 4));
     } else {
         k_seg_sum_s<<<nblocks(N * D, BLOCK), BLOCK, 0, s>>>(
             msg, row_ptr, base, out, N, (int32_t)D);
@@ -725,7 +732,11 @@ int dm_seg_sum_gather_f32(const float* msg, const int32_t* perm,
                           float* out, int64_t N, int64_t D, uint64_t stream) {
     hipStream_t s = (hipStream_t)stream;
     if (D % 4 == 0 && D / 4 <= 16) {
-        k_seg_sum_gather_v4<<<nblocks(N, 16), BLOCK, 0, s>>>(
+        k_seg_sum_gather_v4<4><<<nblocks(N, 16), BLOCK, 0, s>>>(
+            (const float4*)msg, perm, row_ptr, (const float4*)base,
+            (float4*)out, N, (int32_t)(D / 4));
+    } else if (D % 4 == 0 && D / 4 <= 32) {
+        k_seg_sum_gather_v4<5><<<nblocks(N, 8), BLOCK, 0, s>>>(
             (const float4*)msg, perm, row_ptr, (const float4*)base,
             (float4*)out, N, (int32_t)(D / 4));
     } else {

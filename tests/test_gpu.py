@@ -45,7 +45,7 @@ class TestKernels:
 
     def test_seg_sum_vs_index_add(self):
         from distmlip_amd.ops import raw_seg_sum
-        for D in (64, 3, 9):
+        for D in (64, 128, 3, 9):
             N, E = 400, 5000
             idx64 = torch.sort(torch.randint(0, N, (E,))).values
             msg = torch.randn(E, D, device=self.dev)
