@@ -347,10 +347,23 @@ class SpmdEngine:
 
         # Verlet-skin mode: the edge list is a superset built at
         # cutoff+skin; mask the shared message weights by the TRUE cutoffs
-        # so every out-of-range contribution is an exact zero
+        # so every out-of-range contribution is an exact zero.
+        # The mask distances are recomputed in FP64 from the same
+        # frac/lattice/offset inputs the builder uses: a fresh build
+        # selects edges/bonds by fp64 `d < cutoff`, and an fp32 distance
+        # flips SELECTION-boundary members (weight layers carry a bias,
+        # so a flipped 3 A bond shifts forces O(1) — 54 flipped atoms at
+        # li100k, run 34 forensics).
         edge_mask = None
+        d64e = None
         if mask_cutoffs:
-            edge_mask = (bond_dist.detach() < cfg.cutoff).to(ft).unsqueeze(1)
+            with torch.no_grad():
+                lat64 = lattice.detach().double()
+                pos64 = frac_local.double() @ lat64
+                off64 = off_local.double() @ lat64
+                bv64 = (pos64[pd.dst.long()] + off64 - pos64[pd.src.long()])
+                d64e = torch.linalg.norm(bv64, dim=1)
+            edge_mask = (d64e < cfg.cutoff).to(ft).unsqueeze(1)
 
         v = core.atom_embedding(species_local)
         e = core.bond_embedding(bond_expansion)
@@ -370,7 +383,12 @@ class SpmdEngine:
 
             bond_mask = None
             if mask_cutoffs:
-                bond_mask = (nd_dist.detach() <
+                with torch.no_grad():
+                    nd_d64 = torch.empty(pd.n_bonds, dtype=torch.float64,
+                                         device=dev).index_copy(
+                        0, pd.map_ude, d64e[pd.map_de])
+                    nd_d64 = _halo(nd_d64.unsqueeze(1), line_plan).squeeze(1)
+                bond_mask = (nd_d64 <
                              cfg.three_body_cutoff).to(ft).unsqueeze(1)
             exp3 = ops.rbf_env(nd_dist, core.rbf_freq_bond,
                                cfg.three_body_cutoff, cfg.cutoff_exponent)
