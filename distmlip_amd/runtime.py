@@ -358,8 +358,15 @@ class SpmdEngine:
         d64e = None
         if mask_cutoffs:
             with torch.no_grad():
-                lat64 = lattice.detach().double()
-                pos64 = frac_local.double() @ lat64
+                # from the ORIGINAL fp64 frac/lattice (the fp32 model
+                # tensors carry ~1e-5 position rounding at a 130 A box —
+                # wider than the selection-flip window)
+                lat64 = torch.tensor(np.asarray(structure.lattice),
+                                     dtype=torch.float64, device=dev)
+                frac64 = torch.tensor(
+                    frac_src if gids is None else frac_src[gids],
+                    dtype=torch.float64, device=dev)
+                pos64 = frac64 @ lat64
                 off64 = off_local.double() @ lat64
                 bv64 = (pos64[pd.dst.long()] + off64 - pos64[pd.src.long()])
                 d64e = torch.linalg.norm(bv64, dim=1)
