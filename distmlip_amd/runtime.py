@@ -63,7 +63,10 @@ def halo_plan(markers, rank: int, P: int) -> List[Tuple[int, int, int, int, int]
 
 import os as _os
 
-_GLOO_HALO_GROUP = None   # emergency fallback: DM_HALO_GLOO=1 stages halo
+_GLOO_HALO_GROUP = None
+
+# selection tolerance of the graph builder (dist.py tol=1e-8 mirror)
+_SEL_TOL = 1e-8   # emergency fallback: DM_HALO_GLOO=1 stages halo
                           # slices through host memory over a gloo group
 
 
@@ -355,7 +358,7 @@ class SpmdEngine:
         # so a flipped 3 A bond shifts forces O(1) — 54 flipped atoms at
         # li100k, run 34 forensics).
         edge_mask = None
-        d64e = None
+        d64sq = None
         if mask_cutoffs:
             with torch.no_grad():
                 # from the ORIGINAL fp64 frac/lattice (the fp32 model
@@ -369,8 +372,13 @@ class SpmdEngine:
                 pos64 = frac64 @ lat64
                 off64 = off_local.double() @ lat64
                 bv64 = (pos64[pd.dst.long()] + off64 - pos64[pd.src.long()])
-                d64e = torch.linalg.norm(bv64, dim=1)
-            edge_mask = (d64e < cfg.cutoff).to(ft).unsqueeze(1)
+                d64sq = (bv64 * bv64).sum(1)
+            # the builder selects by d^2 < r^2 + tol (fpis.c:760-764,
+            # tol=1e-8): replicate the COMPARATOR, not just the radius —
+            # strict `<` would drop exactly-on-boundary members a fresh
+            # build keeps (tests/test_verlet_skin.py boundary test)
+            edge_mask = (d64sq < cfg.cutoff ** 2 + _SEL_TOL).to(
+                ft).unsqueeze(1)
 
         v = core.atom_embedding(species_local)
         e = core.bond_embedding(bond_expansion)
@@ -391,12 +399,13 @@ class SpmdEngine:
             bond_mask = None
             if mask_cutoffs:
                 with torch.no_grad():
-                    nd_d64 = torch.empty(pd.n_bonds, dtype=torch.float64,
-                                         device=dev).index_copy(
-                        0, pd.map_ude, d64e[pd.map_de])
-                    nd_d64 = _halo(nd_d64.unsqueeze(1), line_plan).squeeze(1)
-                bond_mask = (nd_d64 <
-                             cfg.three_body_cutoff).to(ft).unsqueeze(1)
+                    nd_d64sq = torch.empty(pd.n_bonds, dtype=torch.float64,
+                                           device=dev).index_copy(
+                        0, pd.map_ude, d64sq[pd.map_de])
+                    nd_d64sq = _halo(nd_d64sq.unsqueeze(1),
+                                     line_plan).squeeze(1)
+                bond_mask = (nd_d64sq < cfg.three_body_cutoff ** 2
+                             + _SEL_TOL).to(ft).unsqueeze(1)
             exp3 = ops.rbf_env(nd_dist, core.rbf_freq_bond,
                                cfg.three_body_cutoff, cfg.cutoff_exponent)
             theta = compute_theta(

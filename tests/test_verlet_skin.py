@@ -9,7 +9,7 @@ import torch
 
 from distmlip_amd.model import CHGNetCore
 from distmlip_amd.runtime import SpmdEngine
-from distmlip_amd.structures import diamond_si
+from distmlip_amd.structures import Structure, diamond_si, random_cell
 from oracle.chgnet_ref import CpuRefOps, oracle_forward
 from oracle.graph_ref import brute_force_neighbors
 
@@ -41,3 +41,62 @@ def test_step_verlet_matches_fresh_oracle():
 
     assert eng._vcache["rebuilds"] == 1, \
         f"graph was rebuilt {eng._vcache['rebuilds']}x; reuse never engaged"
+
+
+def test_step_verlet_skewed_cell_with_stress():
+    """Masked-superset reuse on a SKEWED cell, with stress, vs the fresh
+    oracle (fp64 exact)."""
+    s = random_cell(260, a=13.5, n_species=3, seed=7, skew=0.07)
+    core = CHGNetCore.seeded(seed=1).double()
+    eng = SpmdEngine(core, world=1, threads=2, device="cpu", ops=CpuRefOps())
+    out = eng.step_verlet(s, skin=0.8, calc_stresses=True)
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    ref = oracle_forward(core, s, g["src"], g["dst"], g["offsets"],
+                         g["within_bond_r"], dtype=torch.float64,
+                         compute_stress=True)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
+    F = np.zeros((s.num_atoms, 3))
+    F[out["global_ids_owned"]] = out["forces_owned"].numpy()
+    assert np.abs(F - ref["forces"].numpy()).max() < 1e-10
+    assert np.abs(out["stress"].numpy() - ref["stress"].numpy()).max() < 1e-8
+
+
+def test_verlet_mask_selection_boundary():
+    """Regression for the run-34 bug class: members sitting EXACTLY on a
+    selection boundary must get the same include/exclude decision from
+    the verlet mask as from a fresh fp64 build.
+
+    Atoms are placed on exact binary fractions of a 12 A cell so the
+    boundary distances (3.0 for the bond graph) are fp64-exact; both
+    sides must apply the builder's `d^2 < r^2 + tol` decision to them.
+    Before the fix the mask re-derived distances from fp32 model tensors
+    with a strict `<`, which flips such members (0.83 eV/A force error
+    at li100k scale, run 34)."""
+    a = 12.0
+    frac = np.array([
+        [0.0, 0.0, 0.0],
+        [0.25, 0.0, 0.0],      # exactly 3.0 A from atom 0 (boundary)
+        [0.0, 0.171875, 0.0],  # 2.0625 A: inside the bond cutoff
+        [0.5, 0.5, 0.5],
+        [0.5, 0.5, 0.171875],  # pair deep in the cell
+    ])
+    s = Structure(frac_coords=frac, lattice=np.eye(3) * a,
+                  species=np.zeros(len(frac), dtype=np.int64),
+                  pbc=np.ones(3, dtype=np.int64))
+    core = CHGNetCore.seeded(seed=0).double()
+    eng = SpmdEngine(core, world=1, threads=2, device="cpu", ops=CpuRefOps())
+    out = eng.step_verlet(s, skin=1.0)
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    # the reference selects by d^2 < r^2 + tol (fpis.c:760-764), so the
+    # exactly-on-boundary pair IS in the fresh three-body set — the mask
+    # must reach the same decision
+    within_d = np.linalg.norm(
+        (s.frac_coords[g["dst"]] + g["offsets"] - s.frac_coords[g["src"]])
+        @ s.lattice, axis=1)[g["within_bond_r"]]
+    assert np.any(np.isclose(within_d, 3.0, atol=1e-12))
+    ref = oracle_forward(core, s, g["src"], g["dst"], g["offsets"],
+                         g["within_bond_r"], dtype=torch.float64)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
+    F = np.zeros((s.num_atoms, 3))
+    F[out["global_ids_owned"]] = out["forces_owned"].numpy()
+    assert np.abs(F - ref["forces"].numpy()).max() < 1e-10
