@@ -189,16 +189,6 @@ def raw_silu_bwd(go_h, go_z, z):
     return dz
 
 
-def raw_silu_bwd_packed(go_p, go_z, z):
-    """Same, with go_p in the fused edge-MLP's packed [2,E,half] h layout
-    (z and the returned dz are row-major [E,2*half])."""
-    dz = torch.empty_like(z)
-    _check(hip_lib().dm_silu_bwd_packed_f32(
-        _fp(go_p), _fp(go_z) if go_z is not None else None, _fp(z), _fp(dz),
-        z.shape[0], z.shape[1] // 2, _stream()), "dm_silu_bwd_packed_f32")
-    return dz
-
-
 class _GatherAdd3(torch.autograd.Function):
     """Emits (z, silu(z)) in one kernel; z doubles as the saved activation
     for the fused silu backward."""
