@@ -555,6 +555,62 @@ class HipOps:
     def scatter_edges(self, msg, pd, base=None):
         return _SegSum.apply(msg, pd.dst, pd.row_ptr, pd.n_atoms, base)
 
+    # -- raw (non-differentiable) primitives for the hand-sequenced conv
+    #    backward (distmlip_amd.conv); see ops_base docstring ------------
+
+    def r_gather_add3(self, zs, zd, ze, pd):
+        _chk_f32(zs, zd, ze)
+        z = torch.empty_like(ze)
+        h = torch.empty_like(ze)
+        _check(hip_lib().dm_gather_add3_f32(
+            _fp(zs), _fp(zd), _fp(ze), _ip(pd.src), _ip(pd.dst), _fp(z),
+            _fp(h), ze.shape[0], ze.shape[1], _stream()),
+            "dm_gather_add3_f32")
+        return z, h
+
+    def _cg_ptrs(self, cg):
+        half = cg.shape[1] * cg.shape[2]
+        return (ctypes.cast(cg.data_ptr(), POINTER(c_float)),
+                ctypes.cast(cg.data_ptr() + 4 * half, POINTER(c_float)),
+                half)
+
+    def r_combine_fwd(self, cg, w, base):
+        _chk_f32(cg, w, base)
+        c_ptr, g_ptr, half = self._cg_ptrs(cg)
+        out = torch.empty(cg.shape[1], cg.shape[2], dtype=cg.dtype,
+                          device=cg.device)
+        _check(hip_lib().dm_gated_combine_fwd_f32(
+            c_ptr, g_ptr, _fp(w) if w is not None else None,
+            _fp(base) if base is not None else None, _fp(out), half,
+            _stream()), "dm_gated_combine_fwd_f32")
+        return out
+
+    def r_combine_bwd(self, go, cg, w):
+        _chk_f32(go, cg, w)
+        c_ptr, g_ptr, half = self._cg_ptrs(cg)
+        dcg = torch.empty_like(cg)
+        dc_ptr = ctypes.cast(dcg.data_ptr(), POINTER(c_float))
+        dg_ptr = ctypes.cast(dcg.data_ptr() + 4 * half, POINTER(c_float))
+        dw = torch.empty_like(w) if w is not None else None
+        _check(hip_lib().dm_gated_combine_bwd_f32(
+            _fp(go), c_ptr, g_ptr, _fp(w) if w is not None else None,
+            dc_ptr, dg_ptr, _fp(dw) if dw is not None else None, half,
+            _stream()), "dm_gated_combine_bwd_f32")
+        return dcg, dw
+
+    def r_silu_bwd(self, go_h, z):
+        return raw_silu_bwd(go_h.contiguous(), None, z)
+
+    def r_gather_dst(self, x, pd):
+        return raw_gather(x.contiguous(), pd.dst)
+
+    def r_seg_dst(self, msg, pd, base=None):
+        return raw_seg_sum(msg, pd.row_ptr, pd.n_atoms, base)
+
+    def r_seg_src(self, msg, pd):
+        return raw_seg_sum_gather(msg, pd.src_perm, pd.src_row_ptr,
+                                  pd.n_atoms)
+
     def scatter_lines(self, msg, pd, base=None):
         return _SegSum.apply(msg, pd.l_dst, pd.line_row_ptr, pd.n_bonds, base)
 

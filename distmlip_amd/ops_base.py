@@ -26,6 +26,20 @@ the permutation CSRs the graph builder emits; `csr` is an optional
                                          -> the 3-gather line-graph form
     scatter_edges(msg, pd, base)         -> base + segment-sum of the
                                             dst-sorted msg rows per node
+
+  RAW primitives (prefix r_): NON-differentiable single kernels used by
+  the hand-sequenced conv backward (distmlip_amd.conv._AtomConvFn),
+  which replaces autograd's per-tensor gradient accumulation for the
+  atom-conv blocks.  Both backends implement them so the CPU fp64
+  exactness tests cover the product's hand-written reverse pass:
+
+    r_gather_add3(zs, zd, ze, pd)        -> (z, silu(z)) raw forward
+    r_combine_fwd(cg, w, base)           -> base + silu(cg0)*sigmoid(cg1)*w
+    r_combine_bwd(go, cg, w)             -> (dcg [2,*,D], dw)
+    r_silu_bwd(go_h, z)                  -> go_h * silu'(z)
+    r_gather_dst(x, pd)                  -> x[pd.dst]
+    r_seg_dst(msg, pd, base)             -> segment-sum over the dst CSR
+    r_seg_src(msg, pd)                   -> permuted segment-sum (src CSR)
     scatter_lines(msg, pd, base)         -> same over the line CSR per bond
     gated_combine(c, g, w, base)         -> base + silu(c)*sigmoid(g)*w
     gated_combine_packed(cg, w, base)    -> same over PACKED cg [2,*,D]
@@ -64,6 +78,20 @@ class OpsBackend(Protocol):
     def rbf_env(self, d, freqs, cutoff, pexp) -> torch.Tensor: ...
 
     def scatter_edges(self, msg, pd, base=None) -> torch.Tensor: ...
+
+    def r_gather_add3(self, zs, zd, ze, pd): ...
+
+    def r_combine_fwd(self, cg, w, base) -> torch.Tensor: ...
+
+    def r_combine_bwd(self, go, cg, w): ...
+
+    def r_silu_bwd(self, go_h, z) -> torch.Tensor: ...
+
+    def r_gather_dst(self, x, pd) -> torch.Tensor: ...
+
+    def r_seg_dst(self, msg, pd, base=None) -> torch.Tensor: ...
+
+    def r_seg_src(self, msg, pd) -> torch.Tensor: ...
 
     def scatter_lines(self, msg, pd, base=None) -> torch.Tensor: ...
 

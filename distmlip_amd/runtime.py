@@ -32,8 +32,10 @@ import numpy as np
 import torch
 import torch.distributed as dist
 
+from distmlip_amd.conv import _AtomConvFn, conv_fn_available
 from distmlip_amd.chgnet import (
     PartitionData,
+    _packed_weights,
     gated_mlp_split3,
     gated_mlp_split4,
 )
@@ -448,6 +450,14 @@ class SpmdEngine:
                 _m(core.bond_bond_weights(bexp), edge_mask)
             wab = w_ab if w_ab is not None else \
                 _m(core.atom_bond_weights(bexp), edge_mask)
+            pe = _packed_weights(blk.edge_mlp)
+            if d % 2 == 0 and conv_fn_available(ops, pe):
+                # recording path: one Function per block, hand-sequenced
+                # reverse pass (distmlip_amd/conv.py)
+                pn = _packed_weights(blk.node_mlp)
+                return _AtomConvFn.apply(v, e, wbb.contiguous(),
+                                         wab.contiguous(), pd, ops,
+                                         pe[:4], pn[:4], d)
             e = gated_mlp_split3(blk.edge_mlp, v, e, pd, ops, d,
                                  w=wbb, base=e)
             msg = gated_mlp_split3(blk.node_mlp, v, e, pd, ops, d, w=wab)

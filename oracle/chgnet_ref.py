@@ -228,6 +228,47 @@ class CpuRefOps:
         return torch.nn.functional.silu(
             arow @ wt + bias + z1[pd.l_src] + z2[pd.l_dst] + zv[pd.center])
 
+    # raw primitives for the hand-sequenced conv backward (ops_base)
+
+    def r_gather_add3(self, zs, zd, ze, pd):
+        z = zs[pd.src] + zd[pd.dst] + ze
+        return z, torch.nn.functional.silu(z)
+
+    def r_combine_fwd(self, cg, w, base):
+        out = torch.nn.functional.silu(cg[0]) * torch.sigmoid(cg[1])
+        if w is not None:
+            out = out * w
+        return out if base is None else base + out
+
+    def r_combine_bwd(self, go, cg, w):
+        c, g = cg[0], cg[1]
+        sc = torch.sigmoid(c)
+        silu_c = c * sc
+        sg = torch.sigmoid(g)
+        gw = go if w is None else go * w
+        dc = gw * sg * (sc * (1 + c * (1 - sc)))
+        dg = gw * silu_c * sg * (1 - sg)
+        dw = go * silu_c * sg if w is not None else None
+        return torch.stack([dc, dg]), dw
+
+    def r_silu_bwd(self, go_h, z):
+        s = torch.sigmoid(z)
+        return go_h * (s * (1 + z * (1 - s)))
+
+    def r_gather_dst(self, x, pd):
+        return x[pd.dst]
+
+    def r_seg_dst(self, msg, pd, base=None):
+        out = torch.zeros((pd.n_atoms,) + tuple(msg.shape[1:]),
+                          dtype=msg.dtype, device=msg.device
+                          ).index_add_(0, pd.dst, msg)
+        return out if base is None else base + out
+
+    def r_seg_src(self, msg, pd):
+        return torch.zeros((pd.n_atoms,) + tuple(msg.shape[1:]),
+                           dtype=msg.dtype, device=msg.device
+                           ).index_add_(0, pd.src, msg)
+
     def edge_geom_rbf(self, pos, offshift, freqs, cutoff, pexp, pd):
         from distmlip_amd.model import bond_expansion_from_dist
         bv = pos[pd.dst] + offshift - pos[pd.src]
