@@ -100,3 +100,27 @@ def test_verlet_mask_selection_boundary():
     F = np.zeros((s.num_atoms, 3))
     F[out["global_ids_owned"]] = out["forces_owned"].numpy()
     assert np.abs(F - ref["forces"].numpy()).max() < 1e-10
+
+
+def test_hand_sequenced_conv_matches_default():
+    """DM_FUSED_CONV=1 (hand-sequenced atom-conv reverse pass,
+    distmlip_amd/conv.py) must match the op-by-op autograd path exactly
+    in fp64 — pins the hand-written backward against the default."""
+    import os
+
+    s = diamond_si((10, 2, 2), jitter=0.1, seed=3)
+    core = CHGNetCore.seeded(seed=0).double()
+    outs = {}
+    for flag in ("0", "1"):
+        os.environ["DM_FUSED_CONV"] = flag
+        try:
+            eng = SpmdEngine(core, world=1, threads=2, device="cpu",
+                             ops=CpuRefOps())
+            outs[flag] = eng.step(s, calc_stresses=True)
+        finally:
+            os.environ.pop("DM_FUSED_CONV", None)
+    assert abs(outs["0"]["energy"].item() - outs["1"]["energy"].item()) < 1e-11
+    dF = (outs["0"]["forces_owned"] - outs["1"]["forces_owned"]).abs().max()
+    assert dF.item() < 1e-11, dF
+    dS = (outs["0"]["stress"] - outs["1"]["stress"]).abs().max()
+    assert dS.item() < 1e-9, dS
