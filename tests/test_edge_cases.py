@@ -91,3 +91,38 @@ def test_straddle_warning_does_not_crash(capfd):
     gids = np.sort(np.concatenate([
         np.asarray(out[4][p])[:out[2][p][3]] for p in range(2)]))
     assert (gids == np.arange(s.num_atoms)).all()
+
+
+def test_engine_empty_bond_graph():
+    """Engine E+F on a structure whose three-body set is EMPTY (simple
+    cubic, nearest neighbor 4 A: inside the 6 A edge cutoff, outside the
+    3 A bond cutoff) — the bond-graph branches must degrade cleanly and
+    match the oracle."""
+    import torch
+
+    from distmlip_amd.model import CHGNetCore
+    from distmlip_amd.runtime import SpmdEngine
+    from distmlip_amd.structures import Structure
+    from oracle.chgnet_ref import CpuRefOps, oracle_forward
+    from oracle.graph_ref import brute_force_neighbors
+
+    rng = np.random.default_rng(1)
+    n = 4
+    cells = np.stack(np.meshgrid(*[np.arange(n)] * 3, indexing="ij"),
+                     -1).reshape(-1, 3)
+    frac = (cells + 0.5) / n + rng.normal(0, 0.004, (n ** 3, 3))
+    s = Structure(frac_coords=np.mod(frac, 1.0),
+                  lattice=np.eye(3) * (4.0 * n),
+                  species=np.zeros(n ** 3, dtype=np.int64),
+                  pbc=np.ones(3, dtype=np.int64))
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    assert len(g["within_bond_r"]) == 0          # the scenario premise
+    core = CHGNetCore.seeded(seed=0).double()
+    eng = SpmdEngine(core, world=1, threads=2, device="cpu", ops=CpuRefOps())
+    out = eng.step(s)
+    ref = oracle_forward(core, s, g["src"], g["dst"], g["offsets"],
+                         g["within_bond_r"], dtype=torch.float64)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
+    F = np.zeros((s.num_atoms, 3))
+    F[out["global_ids_owned"]] = out["forces_owned"].numpy()
+    assert np.abs(F - ref["forces"].numpy()).max() < 1e-10
