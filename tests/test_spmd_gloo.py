@@ -11,7 +11,8 @@ import torch.distributed as dist
 import torch.multiprocessing as mp
 
 
-def _worker(rank, world, init_file, out_dir, calc_stresses, checkpoint="auto"):
+def _worker(rank, world, init_file, out_dir, calc_stresses, checkpoint="auto",
+            verlet=False):
     import sys
     sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     from distmlip_amd.model import CHGNetCore
@@ -22,11 +23,18 @@ def _worker(rank, world, init_file, out_dir, calc_stresses, checkpoint="auto"):
     dist.init_process_group("gloo", init_method=f"file://{init_file}",
                             rank=rank, world_size=world)
     try:
-        s = diamond_si((12, 2, 2), jitter=0.12, seed=2)
+        # world=4 needs slabs wider than 2*(cutoff+bond_cutoff)=18 A
+        reps = (16, 2, 2) if world >= 4 else (12, 2, 2)
+        s = diamond_si(reps, jitter=0.12, seed=2)
         core = CHGNetCore.seeded(seed=0).double()
         eng = SpmdEngine(core, world, threads=2, device="cpu", ops=CpuRefOps(),
                          checkpoint=checkpoint)
-        out = eng.step(s, calc_stresses=calc_stresses)
+        if verlet:
+            # masked-superset path under SPMD: exercises the fp64 mask
+            # halo (nd_d64sq) across ranks
+            out = eng.step_verlet(s, skin=0.8, calc_stresses=calc_stresses)
+        else:
+            out = eng.step(s, calc_stresses=calc_stresses)
         np.save(os.path.join(out_dir, f"E_{rank}.npy"),
                 np.array([out["energy"].item()]))
         np.save(os.path.join(out_dir, f"F_{rank}.npy"),
@@ -40,13 +48,16 @@ def _worker(rank, world, init_file, out_dir, calc_stresses, checkpoint="auto"):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world,calc_stresses,checkpoint", [
-    (2, False, "auto"),
-    (2, True, "auto"),
-    (3, False, "auto"),
-    (2, False, "on"),     # forced activation checkpointing (1M-atom path)
+@pytest.mark.parametrize("world,calc_stresses,checkpoint,verlet", [
+    (2, False, "auto", False),
+    (2, True, "auto", False),
+    (3, False, "auto", False),
+    (2, False, "on", False),  # forced activation checkpointing (1M path)
+    (4, False, "auto", False),  # the driver's 8-GPU shape, scaled down
+    (2, True, "auto", True),    # verlet masked-superset across ranks
 ])
-def test_spmd_ranks_match_oracle(world, calc_stresses, checkpoint, tmp_path):
+def test_spmd_ranks_match_oracle(world, calc_stresses, checkpoint, verlet,
+                                 tmp_path):
     from distmlip_amd.model import CHGNetCore
     from distmlip_amd.structures import diamond_si
     from oracle.chgnet_ref import oracle_forward
@@ -55,10 +66,11 @@ def test_spmd_ranks_match_oracle(world, calc_stresses, checkpoint, tmp_path):
     init_file = str(tmp_path / "pg_init")
     out_dir = str(tmp_path)
     mp.spawn(_worker, args=(world, init_file, out_dir, calc_stresses,
-                            checkpoint),
+                            checkpoint, verlet),
              nprocs=world, join=True)
 
-    s = diamond_si((12, 2, 2), jitter=0.12, seed=2)
+    s = diamond_si((16, 2, 2) if world >= 4 else (12, 2, 2),
+                   jitter=0.12, seed=2)
     g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
     core = CHGNetCore.seeded(seed=0).double()
     ref = oracle_forward(core, s, g["src"], g["dst"], g["offsets"],
