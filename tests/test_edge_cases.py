@@ -126,3 +126,34 @@ def test_engine_empty_bond_graph():
     F = np.zeros((s.num_atoms, 3))
     F[out["global_ids_owned"]] = out["forces_owned"].numpy()
     assert np.abs(F - ref["forces"].numpy()).max() < 1e-10
+
+
+def test_oracle_rotation_covariance():
+    """Oracle physics sanity: rotating the cell leaves the energy
+    invariant and rotates forces covariantly (F' = F @ R).  Strengthens
+    the oracle pinning: a wrong geometry chain (bond_vec/theta) would
+    break this."""
+    import torch
+    from scipy.stats import ortho_group
+
+    from distmlip_amd.model import CHGNetCore
+    from distmlip_amd.structures import random_cell
+    from oracle.chgnet_ref import oracle_forward
+    from oracle.graph_ref import brute_force_neighbors
+
+    s = random_cell(40, a=11.0, n_species=2, seed=3, skew=0.05)
+    core = CHGNetCore.seeded(seed=0)
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    ref = oracle_forward(core, s, g["src"], g["dst"], g["offsets"],
+                         g["within_bond_r"], dtype=torch.float64)
+
+    R = ortho_group.rvs(3, random_state=11)
+    s2 = type(s)(frac_coords=s.frac_coords, lattice=s.lattice @ R,
+                 species=s.species, pbc=s.pbc)
+    g2 = brute_force_neighbors(s2.frac_coords, s2.lattice, s2.pbc, 6.0, 3.0)
+    ref2 = oracle_forward(core, s2, g2["src"], g2["dst"], g2["offsets"],
+                          g2["within_bond_r"], dtype=torch.float64)
+
+    assert abs(ref["energy"].item() - ref2["energy"].item()) < 1e-9
+    FR = ref["forces"].numpy() @ R
+    assert np.abs(FR - ref2["forces"].numpy()).max() < 1e-9
