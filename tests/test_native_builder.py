@@ -108,3 +108,32 @@ def test_single_partition_allowed():
     assert len(out[0]) == 1
     assert out[2][0].tolist() == [0, s.num_atoms, s.num_atoms]
     assert len(out[0][0]) == len(out[5])   # all edges local
+
+
+@pytest.mark.parametrize("seed", range(12))
+def test_native_vs_bruteforce_randomized(seed):
+    """Randomized breadth: random atom counts, cell sizes, skews and
+    species — native builder edge set + distances must match the O(N^2)
+    brute force exactly (the fixed-structure tests' property, fuzzed)."""
+    rng = np.random.default_rng(1000 + seed)
+    n = int(rng.integers(2, 60))
+    a = float(rng.uniform(12.5, 18.0))
+    skew = float(rng.uniform(0.0, 0.12))
+    s = random_cell(n, a=a, n_species=3, seed=int(rng.integers(1 << 30)),
+                    skew=skew)
+    ours = capi.get_subgraphs_fast(s.cart_coords, 6.0, s.pbc, s.lattice, 1,
+                                   3.0, 1e-8, 2, True, s.frac_coords)
+    bf = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    k1 = edge_key(ours[5], ours[6], ours[7])
+    o1 = canonical_edge_order(ours[5], ours[6], ours[7])
+    k2 = edge_key(bf["src"], bf["dst"], bf["offsets"])
+    o2 = canonical_edge_order(bf["src"], bf["dst"], bf["offsets"])
+    assert k1.shape == k2.shape, (seed, n, a, skew)
+    assert (k1[o1] == k2[o2]).all(), (seed, n, a, skew)
+    assert np.abs(np.asarray(ours[8])[o1] - bf["dist"][o2]).max() < 1e-12
+    # three-body subset agrees as sets of canonical edge keys
+    wi = np.asarray(ours[11], dtype=np.int64).ravel()   # g["within"]
+    w2i = np.asarray(bf["within_bond_r"]).ravel()
+    w1 = set(map(tuple, k1[wi].reshape(len(wi), 5).tolist())) if len(wi) else set()
+    w2 = set(map(tuple, k2[w2i].reshape(len(w2i), 5).tolist())) if len(w2i) else set()
+    assert w1 == w2, (seed, n, a, skew)
