@@ -6,7 +6,7 @@ import numpy as np
 import pytest
 
 from distmlip_amd import capi
-from distmlip_amd.structures import diamond_si, random_cell
+from distmlip_amd.structures import Structure, diamond_si, random_cell
 from oracle.graph_ref import brute_force_neighbors, canonical_edge_order, edge_key
 
 
@@ -137,3 +137,35 @@ def test_native_vs_bruteforce_randomized(seed):
     w1 = set(map(tuple, k1[wi].reshape(len(wi), 5).tolist())) if len(wi) else set()
     w2 = set(map(tuple, k2[w2i].reshape(len(w2i), 5).tolist())) if len(w2i) else set()
     assert w1 == w2, (seed, n, a, skew)
+
+
+@pytest.mark.parametrize("seed", range(6))
+def test_partitioned_fuzz_covers_global_graph(seed):
+    """Fuzzed P=2/3 partition builds on random elongated cells: owned
+    edges across partitions must union EXACTLY to the global edge set
+    (each edge owned by its dst's partition, once)."""
+    rng = np.random.default_rng(2000 + seed)
+    P = int(rng.integers(2, 4))
+    n = int(rng.integers(40, 120))
+    # elongated box so slabs pass the width check: long axis > P*18
+    long_a = float(P * 19 + rng.uniform(0, 8))
+    lat = np.diag([long_a, 13.0, 13.5])
+    frac = rng.random((n, 3))
+    s = Structure(frac_coords=frac, lattice=lat,
+                  species=np.zeros(n, dtype=np.int64),
+                  pbc=np.ones(3, dtype=np.int64))
+    ours, parts = capi.get_subgraphs_fast(
+        s.cart_coords, 6.0, s.pbc, s.lattice, P, 3.0, 1e-8, 2, True,
+        s.frac_coords, return_csr=True)
+    bf = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    kg = edge_key(bf["src"], bf["dst"], bf["offsets"])
+    gset = set(map(tuple, kg.tolist()))
+    seen = set()
+    for p in range(P):
+        egids = np.asarray(ours[16][p], dtype=np.int64)  # per-edge gids
+        kp = edge_key(np.asarray(ours[5])[egids], np.asarray(ours[6])[egids],
+                      np.asarray(ours[7])[egids])
+        for t in map(tuple, kp.tolist()):
+            assert t not in seen, f"edge owned twice (seed {seed})"
+            seen.add(t)
+    assert seen == gset, (seed, P, n, len(seen), len(gset))
