@@ -6,13 +6,29 @@ import numpy as np
 import pytest
 
 from distmlip_amd import capi
-from distmlip_amd.structures import diamond_si, random_cell
+from distmlip_amd.structures import Structure, diamond_si, random_cell
+
+
+def _fuzz_case(seed):
+    """Random elongated box sized for the slab width check."""
+    rng = np.random.default_rng(3000 + seed)
+    P = int(rng.integers(2, 4))
+    n = int(rng.integers(60, 160))
+    long_a = float(P * 19 + rng.uniform(0, 10))
+    lat = np.diag([long_a, 13.0, 14.0])
+    frac = rng.random((n, 3))
+    return Structure(frac_coords=frac, lattice=lat,
+                     species=np.zeros(n, dtype=np.int64),
+                     pbc=np.ones(3, dtype=np.int64)), P
 
 
 CASES = [
     ("si_12x2x2_P2", lambda: diamond_si((12, 2, 2), jitter=0.12, seed=2), 2),
     ("si_12x2x2_P3", lambda: diamond_si((12, 2, 2), jitter=0.12, seed=2), 3),
     ("rand200_skew", lambda: random_cell(200, a=40.0, seed=5, skew=0.05), 2),
+] + [
+    (f"fuzz{k}", (lambda k=k: _fuzz_case(k)[0]), _fuzz_case(k)[1])
+    for k in range(6)
 ]
 
 TUPLE_FIELDS = {0: "src_local", 1: "dst_local", 2: "markers",
