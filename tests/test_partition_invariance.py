@@ -60,3 +60,33 @@ def test_stress_invariance_fp64(core64, si_slab, si_slab_graph, ref_graph_backen
                           calc_stresses=True)
     assert S is not None
     assert (S - ref["stress"]).abs().max().item() < 1e-10
+
+
+@pytest.mark.parametrize("seed", range(3))
+def test_engine_vs_oracle_randomized(seed):
+    """Randomized engine-vs-oracle breadth on CPU fp64: random skewed
+    cells with mixed species through SpmdEngine world=1."""
+    import torch
+
+    from distmlip_amd.runtime import SpmdEngine
+    from distmlip_amd.structures import random_cell
+    from oracle.chgnet_ref import CpuRefOps, oracle_forward
+    from oracle.graph_ref import brute_force_neighbors
+
+    rng = np.random.default_rng(4000 + seed)
+    s = random_cell(int(rng.integers(60, 160)),
+                    a=float(rng.uniform(12.0, 16.0)), n_species=4,
+                    seed=int(rng.integers(1 << 30)),
+                    skew=float(rng.uniform(0, 0.08)))
+    core = CHGNetCore.seeded(seed=seed).double()
+    eng = SpmdEngine(core, world=1, threads=2, device="cpu", ops=CpuRefOps())
+    out = eng.step(s, calc_stresses=True)
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 3.0)
+    ref = oracle_forward(core, s, g["src"], g["dst"], g["offsets"],
+                         g["within_bond_r"], dtype=torch.float64,
+                         compute_stress=True)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
+    F = np.zeros((s.num_atoms, 3))
+    F[out["global_ids_owned"]] = out["forces_owned"].numpy()
+    assert np.abs(F - ref["forces"].numpy()).max() < 1e-10, seed
+    assert np.abs(out["stress"].numpy() - ref["stress"].numpy()).max() < 1e-8
