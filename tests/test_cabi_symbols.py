@@ -41,3 +41,15 @@ def test_ctypes_loads_without_package():
             pytest.skip(f"{libname} not built")
         lib = ctypes.CDLL(path)
         assert lib is not None
+
+
+def test_bench_cpu_baseline_leg_runs():
+    """The bench contract's cpu_baseline leg (oracle timed on host cores)
+    must stay runnable without a GPU and report the declared fields."""
+    import bench
+    out = bench.cpu_baseline_leg("li100k", 4)
+    assert out["kind"] == "port"
+    assert out["unit"] == "atom_steps_per_s"
+    assert out["value"] > 0
+    assert out["cores"] == 4
+    assert "sample" in out
