@@ -118,3 +118,51 @@ def conv_fn_available(ops, mlp_pack) -> bool:
             and hasattr(ops, "r_gather_add3")
             and not mlp_pack[0].requires_grad
             and torch.is_grad_enabled())
+
+
+class _BondConvFn(torch.autograd.Function):
+    """(n, a, v, w3) -> n' for one bond (line-graph) conv, hand-written
+    backward (reference chgnet.py:326-368 semantics as orchestrated in
+    runtime.bond_body).  mask_l ([L,1] detached, or None) is the verlet
+    destination-bond mask applied to the line weights."""
+
+    @staticmethod
+    def forward(ctx, n, a, v, w3, pd, ops, packs, d, mask_l):
+        wcg, bcg, w2, b2 = packs
+        w1, wn2, wa, wv = (wcg[:, :d], wcg[:, d:2 * d], wcg[:, 2 * d:3 * d],
+                           wcg[:, 3 * d:])
+        wl = ops.r_gather_lsrc(w3, pd)
+        if mask_l is not None:
+            wl = wl * mask_l
+        z, h = ops.r_gather_add4(n @ w1.t(), n @ wn2.t(),
+                                 torch.addmm(bcg, a, wa.t()), v @ wv.t(), pd)
+        cg = _second_fwd(h, w2, b2, d)
+        msg = ops.r_combine_fwd(cg, wl, None)
+        n2 = ops.r_seg_ldst(msg, pd, base=n)
+        ctx.set_materialize_grads(False)
+        ctx.save_for_backward(z, cg, wl)
+        ctx.pd, ctx.ops, ctx.d = pd, ops, d
+        ctx.packs = (w1, wn2, wa, wv, w2)
+        ctx.has_mask = mask_l is not None
+        ctx.mask_l = mask_l
+        return n2
+
+    @staticmethod
+    def backward(ctx, go_n2):
+        z, cg, wl = ctx.saved_tensors
+        pd, ops, d = ctx.pd, ctx.ops, ctx.d
+        w1, wn2, wa, wv, w2 = ctx.packs
+        with torch.no_grad():
+            go_n2 = go_n2.contiguous()
+            dmsg = ops.r_gather_ldst(go_n2, pd)
+            dcg, dwl = ops.r_combine_bwd(dmsg, cg, wl)
+            dz = ops.r_silu_bwd(_second_bwd(dcg, w2, d), z)
+            da = dz @ wa
+            # n grad: base passthrough + both per-bond GEMM backs, fused
+            gn = torch.addmm(go_n2, ops.r_seg_lsrc(dz, pd), w1)
+            gn.addmm_(ops.r_seg_ldst(dz, pd), wn2)
+            gv = ops.r_seg_center(dz, pd) @ wv
+            if ctx.has_mask:
+                dwl = dwl * ctx.mask_l
+            gw3 = ops.r_seg_lsrc(dwl, pd)
+        return gn, da, gv, gw3, None, None, None, None, None

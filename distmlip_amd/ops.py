@@ -611,6 +611,35 @@ class HipOps:
         return raw_seg_sum_gather(msg, pd.src_perm, pd.src_row_ptr,
                                   pd.n_atoms)
 
+    # line-graph raw primitives (bond-conv hand-sequenced reverse)
+
+    def r_gather_add4(self, z1, z2, za, zv, pd):
+        _chk_f32(z1, z2, za, zv)
+        z = torch.empty_like(za)
+        h = torch.empty_like(za)
+        _check(hip_lib().dm_gather_add4_f32(
+            _fp(z1), _fp(z2), _fp(za), _fp(zv), _ip(pd.l_src), _ip(pd.l_dst),
+            _ip(pd.center), _fp(z), _fp(h), za.shape[0], za.shape[1],
+            _stream()), "dm_gather_add4_f32")
+        return z, h
+
+    def r_gather_lsrc(self, x, pd):
+        return raw_gather(x.contiguous(), pd.l_src)
+
+    def r_gather_ldst(self, x, pd):
+        return raw_gather(x.contiguous(), pd.l_dst)
+
+    def r_seg_ldst(self, msg, pd, base=None):
+        return raw_seg_sum(msg, pd.line_row_ptr, pd.n_bonds, base)
+
+    def r_seg_lsrc(self, msg, pd):
+        return raw_seg_sum_gather(msg, pd.line_src_perm, pd.line_src_row_ptr,
+                                  pd.n_bonds)
+
+    def r_seg_center(self, msg, pd):
+        return raw_seg_sum_gather(msg, pd.center_perm, pd.center_row_ptr,
+                                  pd.n_atoms)
+
     def scatter_lines(self, msg, pd, base=None):
         return _SegSum.apply(msg, pd.l_dst, pd.line_row_ptr, pd.n_bonds, base)
 

@@ -32,7 +32,8 @@ import numpy as np
 import torch
 import torch.distributed as dist
 
-from distmlip_amd.conv import _AtomConvFn, conv_fn_available
+from distmlip_amd.conv import (_AtomConvFn, _BondConvFn,
+                               conv_fn_available)
 from distmlip_amd.chgnet import (
     PartitionData,
     _packed_weights,
@@ -65,11 +66,12 @@ def halo_plan(markers, rank: int, P: int) -> List[Tuple[int, int, int, int, int]
 
 import os as _os
 
+# emergency fallback: DM_HALO_GLOO=1 stages halo slices through host
+# memory over a gloo group
 _GLOO_HALO_GROUP = None
 
 # selection tolerance of the graph builder (dist.py tol=1e-8 mirror)
-_SEL_TOL = 1e-8   # emergency fallback: DM_HALO_GLOO=1 stages halo
-                          # slices through host memory over a gloo group
+_SEL_TOL = 1e-8
 
 
 def _exchange(feat: torch.Tensor, plan, reverse: bool = False) -> dict:
@@ -481,6 +483,16 @@ class SpmdEngine:
                 def bond_body(n, a, v, e3, _blk=blk):
                     w3 = w_3b if w_3b is not None else \
                         _m(core.threebody_bond_weights(e3), bond_mask)
+                    pb = _packed_weights(_blk.bond_mlp)
+                    if (d % 2 == 0 and conv_fn_available(ops, pb)
+                            and hasattr(ops, "r_gather_add4")):
+                        mask_l = None
+                        if bond_mask is not None:
+                            mask_l = ops.r_gather_ldst(
+                                bond_mask.contiguous(), pd)
+                        return _BondConvFn.apply(
+                            n, a.contiguous(), v, w3.contiguous(), pd, ops,
+                            pb[:4], d, mask_l)
                     wl = ops.gather(w3, pd.l_src, csr=pd.line_src_csr)
                     if bond_mask is not None:
                         # Verlet mode: a message into a MASKED (superset)

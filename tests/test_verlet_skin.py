@@ -110,17 +110,21 @@ def test_hand_sequenced_conv_matches_default():
 
     s = diamond_si((10, 2, 2), jitter=0.1, seed=3)
     core = CHGNetCore.seeded(seed=0).double()
-    outs = {}
-    for flag in ("0", "1"):
-        os.environ["DM_FUSED_CONV"] = flag
-        try:
-            eng = SpmdEngine(core, world=1, threads=2, device="cpu",
-                             ops=CpuRefOps())
-            outs[flag] = eng.step(s, calc_stresses=True)
-        finally:
-            os.environ.pop("DM_FUSED_CONV", None)
-    assert abs(outs["0"]["energy"].item() - outs["1"]["energy"].item()) < 1e-11
-    dF = (outs["0"]["forces_owned"] - outs["1"]["forces_owned"]).abs().max()
-    assert dF.item() < 1e-11, dF
-    dS = (outs["0"]["stress"] - outs["1"]["stress"]).abs().max()
-    assert dS.item() < 1e-9, dS
+    for verlet in (False, True):       # True: bond masks through the
+        outs = {}                      # hand-sequenced bond conv
+        for flag in ("0", "1"):
+            os.environ["DM_FUSED_CONV"] = flag
+            try:
+                eng = SpmdEngine(core, world=1, threads=2, device="cpu",
+                                 ops=CpuRefOps())
+                outs[flag] = (eng.step_verlet(s, skin=0.8, calc_stresses=True)
+                              if verlet else eng.step(s, calc_stresses=True))
+            finally:
+                os.environ.pop("DM_FUSED_CONV", None)
+        dE = abs(outs["0"]["energy"].item() - outs["1"]["energy"].item())
+        assert dE < 1e-11, (verlet, dE)
+        dF = (outs["0"]["forces_owned"] -
+              outs["1"]["forces_owned"]).abs().max()
+        assert dF.item() < 1e-11, (verlet, dF)
+        dS = (outs["0"]["stress"] - outs["1"]["stress"]).abs().max()
+        assert dS.item() < 1e-9, (verlet, dS)
