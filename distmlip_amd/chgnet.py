@@ -68,7 +68,10 @@ def _packed_weights(mlp: GatedMLP):
       (columns 2d:3d of wcg) transposed for the fused edge-MLP kernel.
     """
     frozen = not any(p.requires_grad for p in mlp.parameters())
-    key = (mlp.core1.weight.device, mlp.core1.weight.dtype)
+    # _version tracks in-place mutation (load_state_dict copy_ bumps it):
+    # a frozen model whose weights are reloaded must not reuse stale packs
+    key = (mlp.core1.weight.device, mlp.core1.weight.dtype,
+           tuple(p._version for p in mlp.parameters()))
     if frozen:
         cached = getattr(mlp, "_dm_packed", None)
         if cached is not None and cached[0] == key:

@@ -162,33 +162,6 @@ def raw_silu_bwd(go_h, go_z, z):
     return dz
 
 
-class _Gather(torch.autograd.Function):
-    @staticmethod
-    def forward(ctx, x, idx, perm, row_ptr):
-        ctx.perm = perm          # None when idx is the sorted direction
-        ctx.row_ptr = row_ptr
-        ctx.n_rows = x.shape[0]
-        return raw_gather(x, idx)
-
-    @staticmethod
-    def backward(ctx, grad):
-        grad = grad.contiguous()
-        if ctx.perm is None:
-            gx = raw_seg_sum(grad, ctx.row_ptr, ctx.n_rows)
-        else:
-            gx = raw_seg_sum_gather(grad, ctx.perm, ctx.row_ptr, ctx.n_rows)
-        return gx, None, None, None
-
-
-def raw_silu_bwd(go_h, go_z, z):
-    """dz = (go_z or 0) + go_h * silu'(z), one fused pass."""
-    dz = torch.empty_like(z)
-    _check(hip_lib().dm_silu_bwd_f32(
-        _fp(go_h), _fp(go_z) if go_z is not None else None, _fp(z), _fp(dz),
-        z.numel(), _stream()), "dm_silu_bwd_f32")
-    return dz
-
-
 class _GatherAdd3(torch.autograd.Function):
     """Emits (z, silu(z)) in one kernel; z doubles as the saved activation
     for the fused silu backward."""

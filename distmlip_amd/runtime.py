@@ -110,17 +110,22 @@ class _HaloSeq:
     """RCCL p2p messages match by POSTING ORDER (no tags): every rank must
     run its halo exchanges in the same order.  Forward order is program
     order; backward order is enforced to be exactly reversed — a violation
-    raises instead of silently crossing buffers."""
-    fwd = 0
-    expect_bwd = None
+    raises instead of silently crossing buffers.  One instance per engine
+    step (created in SpmdEngine.step), so engines in the same process
+    cannot cross-contaminate sequence numbers."""
+
+    def __init__(self):
+        self.fwd = 0
+        self.expect_bwd = None
 
 
 class HaloExchange(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, feat, plan):
+    def forward(ctx, feat, plan, seq_state):
         ctx.plan = plan
-        ctx.seq = _HaloSeq.fwd
-        _HaloSeq.fwd += 1
+        ctx.seq_state = seq_state
+        ctx.seq = seq_state.fwd
+        seq_state.fwd += 1
         recvs = _exchange(feat.detach(), plan, reverse=False)
         out = feat.detach().clone()
         for (q, ss, se, rs, re) in plan:
@@ -134,13 +139,14 @@ class HaloExchange(torch.autograd.Function):
         # sequence-number order, so live exchanges run in strictly
         # DECREASING forward order on every rank (dead branches are skipped
         # identically on all ranks — graph structure is rank-invariant).
-        if _HaloSeq.expect_bwd is None:
-            _HaloSeq.expect_bwd = _HaloSeq.fwd
-        if ctx.seq >= _HaloSeq.expect_bwd:
+        st = ctx.seq_state
+        if st.expect_bwd is None:
+            st.expect_bwd = st.fwd
+        if ctx.seq >= st.expect_bwd:
             raise RuntimeError(
                 f"halo backward out of order: got seq {ctx.seq} after "
-                f"{_HaloSeq.expect_bwd} — rank-divergent exchange order")
-        _HaloSeq.expect_bwd = ctx.seq
+                f"{st.expect_bwd} — rank-divergent exchange order")
+        st.expect_bwd = ctx.seq
         plan = ctx.plan
         g = grad.contiguous()
         recvs = _exchange(g, plan, reverse=True)     # send ghost grads home
@@ -151,7 +157,7 @@ class HaloExchange(torch.autograd.Function):
         for (q, ss, se, rs, re) in plan:
             if se > ss:
                 g[ss:se] += recvs[q]                 # peers' ghost grads
-        return g, None
+        return g, None, None
 
 
 class SpmdEngine:
@@ -248,7 +254,12 @@ class SpmdEngine:
         rebuild = True
         if getattr(self, "_vcache", None) is not None:
             c = self._vcache
-            if c["skin"] == skin and len(c["frac"]) == structure.num_atoms:
+            # a lattice change (NPT / cell rescale) invalidates the stored
+            # edge superset even with unchanged frac coords — compare the
+            # cached lattice, not just displacement
+            if (c["skin"] == skin and len(c["frac"]) == structure.num_atoms
+                    and np.array_equal(c["lattice"],
+                                       np.asarray(structure.lattice))):
                 d = structure.frac_coords - c["frac"]
                 d = (d + 0.5) % 1.0 - 0.5                 # minimal image
                 disp2 = ((d @ structure.lattice) ** 2).sum(1)
@@ -256,6 +267,7 @@ class SpmdEngine:
         if rebuild:
             self._vcache = {
                 "skin": skin,
+                "lattice": np.asarray(structure.lattice).copy(),
                 "frac": structure.frac_coords.copy(),
                 "dist_info": self.build_graph(structure, skin=skin),
                 "rebuilds": getattr(self, "_vcache", None) and
@@ -296,8 +308,7 @@ class SpmdEngine:
                     self.use_bond_graph, dev, frac_override=frac_override)
             else:
                 dist_info = self.build_graph(structure)
-        _HaloSeq.fwd = 0
-        _HaloSeq.expect_bwd = None
+        halo_seq = _HaloSeq()                 # fresh guard per step
         if gpu_pd is not None:
             pd = gpu_pd
             plan, line_plan = [], []
@@ -312,7 +323,7 @@ class SpmdEngine:
             n_owned = dist_info.num_owned_atoms(r)
 
         def _halo(feat, pl):
-            return feat if not pl else HaloExchange.apply(feat, pl)
+            return feat if not pl else HaloExchange.apply(feat, pl, halo_seq)
 
         # ---- local geometry (per-rank; no GPU0 serialization point)
         lat0 = torch.tensor(np.asarray(structure.lattice), dtype=ft, device=dev)

@@ -176,6 +176,7 @@ def test_gated_combine_fwd_bwd():
             assert torch.allclose(base.grad, b2.grad)
 
 
+@requires_gpu
 def test_gated_combine_packed_fwd_bwd():
     """Packed cg [2,E,D] variant matches the separate-tensor op + grads."""
     from distmlip_amd.ops import _GatedCombinePacked
