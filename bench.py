@@ -284,6 +284,14 @@ def main():
     value = total_atoms * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
+    if rank == 0 and os.environ.get("DM_TUNABLEOP_WRITE"):
+        # persist GEMM algorithm choices tuned during warmup so the
+        # shipped tunableop_gfx950.csv can be extended offline
+        try:
+            torch.cuda.tunable.write_file(os.environ["DM_TUNABLEOP_WRITE"])
+        except Exception as e:
+            print(f"# tunableop write failed: {e}", file=sys.stderr)
+
     if rank == 0:
         roofline = timer.summary()
         cpu_b = None
