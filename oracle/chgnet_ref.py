@@ -20,6 +20,13 @@ Model-arithmetic parity vs upstream matgl is UNPINNED (oracle/__init__.py);
 this file IS the executable definition the product must match.  Tests use
 it in fp64 (tight partition-invariance) and fp32 (GPU parity tolerance).
 
+Independence: every FORMULA here is oracle-local (oracle/basis_ref.py —
+radial Bessel / cutoff envelope / Fourier / theta / gated-MLP / linear
+application), restated separately from the product's copies in
+distmlip_amd/model.py; `CHGNetCore` is imported ONLY as the weight
+container (no product forward() code runs here), so a transcription error
+in the product's formulas cannot validate itself.
+
 TEST INFRASTRUCTURE ONLY — see oracle/__init__.py.
 """
 from __future__ import annotations
@@ -27,11 +34,14 @@ from __future__ import annotations
 import numpy as np
 import torch
 
-from distmlip_amd.model import (
-    CHGNetCore,
-    bond_expansion_from_dist,
-    compute_theta,
-    fourier_expansion,
+from distmlip_amd.model import CHGNetCore  # weights container only
+from oracle.basis_ref import (
+    bond_expansion_from_dist_t,
+    compute_theta_t,
+    fourier_expansion_t,
+    gated_mlp,
+    linear,
+    silu,
 )
 
 
@@ -104,14 +114,14 @@ def oracle_forward(core: CHGNetCore, structure, src, dst, offsets, within_idx,
     bond_vec = pos[dst_t] + offshift - pos[src_t]            # chgnet.py:96-99
     bond_dist = torch.linalg.norm(bond_vec, dim=1)           # chgnet.py:100
 
-    bond_expansion = bond_expansion_from_dist(
+    bond_expansion = bond_expansion_from_dist_t(
         bond_dist, core.rbf_freq_atom, cfg.cutoff, cfg.cutoff_exponent)
 
-    v = core.atom_embedding(species)
-    e = core.bond_embedding(bond_expansion)
+    v = core.atom_embedding.weight[species]
+    e = linear(core.bond_embedding, bond_expansion)
 
-    w_ab = core.atom_bond_weights(bond_expansion)
-    w_bb = core.bond_bond_weights(bond_expansion)
+    w_ab = linear(core.atom_bond_weights, bond_expansion)
+    w_bb = linear(core.bond_bond_weights, bond_expansion)
 
     use_bg = cfg.use_bond_graph and len(within_idx) > 0
     if use_bg:
@@ -121,14 +131,15 @@ def oracle_forward(core: CHGNetCore, structure, src, dst, offsets, within_idx,
         l_dst = torch.tensor(l_dst_np, dtype=torch.long)
         center = torch.tensor(center_np, dtype=torch.long)
 
-        bond_expansion3 = bond_expansion_from_dist(
+        bond_expansion3 = bond_expansion_from_dist_t(
             bond_dist[within_t], core.rbf_freq_bond, cfg.three_body_cutoff,
             cfg.cutoff_exponent)
-        w_3b = core.threebody_bond_weights(bond_expansion3)
+        w_3b = linear(core.threebody_bond_weights, bond_expansion3)
 
         bv_b = bond_vec[within_t]
-        theta = compute_theta(bv_b[l_src], bv_b[l_dst])      # chgnet.py:190-194
-        a = core.angle_embedding(fourier_expansion(theta, core.angle_freq))
+        theta = compute_theta_t(bv_b[l_src], bv_b[l_dst])    # chgnet.py:190-194
+        a = linear(core.angle_embedding,
+                   fourier_expansion_t(theta, core.angle_freq))
         n = e[within_t]                                      # chgnet.py:255-261
 
     n_atoms = len(species)
@@ -136,9 +147,9 @@ def oracle_forward(core: CHGNetCore, structure, src, dst, offsets, within_idx,
     def atom_conv(block, v, e):
         """edge update then node update (node sees updated edges)."""
         x = torch.cat([v[src_t], v[dst_t], e], dim=1)
-        e = e + block.edge_mlp(x) * w_bb
+        e = e + gated_mlp(block.edge_mlp, x) * w_bb
         x = torch.cat([v[src_t], v[dst_t], e], dim=1)
-        msg = block.node_mlp(x) * w_ab
+        msg = gated_mlp(block.node_mlp, x) * w_ab
         v = v + torch.zeros_like(v).index_add_(0, dst_t, msg)
         return v, e
 
@@ -148,17 +159,20 @@ def oracle_forward(core: CHGNetCore, structure, src, dst, offsets, within_idx,
             n = e[within_t]                                  # edge_to_bond
             blk = core.bond_convs[layer_i]
             x = torch.cat([n[l_src], n[l_dst], a, v[center]], dim=1)
-            msg = blk.bond_mlp(x) * w_3b[l_src]
+            msg = gated_mlp(blk.bond_mlp, x) * w_3b[l_src]
             n = n + torch.zeros_like(n).index_add_(0, l_dst, msg)
             e = e.index_put((within_t,), n)                  # bond_to_edge
             x = torch.cat([n[l_src], n[l_dst], a, v[center]], dim=1)
-            a = a + blk.angle_mlp(x)
+            a = a + gated_mlp(blk.angle_mlp, x)
 
-    site_props = core.sitewise_readout(v)                    # chgnet.py:391-398
+    site_props = linear(core.sitewise_readout, v)            # chgnet.py:391-398
 
     v, e = atom_conv(core.atom_convs[-1], v, e)              # chgnet.py:400-419
 
-    atom_energies = core.final_layer(v)                      # chgnet.py:422-440
+    # final_layer = Linear-SiLU-Linear-SiLU-Linear (chgnet.py:422-440),
+    # applied with oracle-local arithmetic
+    fl = core.final_layer
+    atom_energies = linear(fl[4], silu(linear(fl[2], silu(linear(fl[0], v)))))
     total_e = atom_energies.sum()
 
     total_e = core.data_std * total_e + core.data_mean       # pes.py:109
