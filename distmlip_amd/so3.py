@@ -262,6 +262,68 @@ def symmetric_basis(nu: int, lo: int) -> np.ndarray:
     return np.stack(basis, axis=-1)
 
 
+@lru_cache(maxsize=None)
+def symmetric_basis_trees(nu: int, lo: int):
+    """Tree factorization of `symmetric_basis(nu, lo)` for cheap
+    contraction with the symmetric power x^(x)nu.
+
+    Returns (trees, M):
+      trees: list of coupling trees — nu=3: (l1, l2, L, l3); nu=2:
+             (l1, l2); nu=1: (lo,)
+      M:     [n_paths, n_trees] with  B_p = sum_t M[p,t] * Sym(T_t).
+    Because x^(x)nu is symmetric, contracting with T_t equals contracting
+    with Sym(T_t), so  phi_p(x) = sum_t M[p,t] * (T_t . x^nu)  — no dense
+    [16]^nu tensors are ever materialized at model run time."""
+    B = symmetric_basis(nu, lo)
+    P = B.shape[-1]
+    trees, sym_flat = [], []
+    if nu == 1:
+        if P:
+            trees.append((lo,))
+            # raw (unnormalized) tree tensor — the runtime contraction
+            # computes plain x-slice / CG contractions, so S columns must
+            # be the raw Sym(T_t), not renormalized
+            T = np.zeros((SH_DIM, 2 * lo + 1))
+            o = L_OFF[lo]
+            T[o:o + L_DIMS[lo], :] = np.eye(L_DIMS[lo])
+            sym_flat.append(T.ravel())
+    elif nu == 2:
+        for l1 in range(L_MAX + 1):
+            for l2 in range(L_MAX + 1):
+                if (l1 + l2 + lo) % 2 == 0 and cg_nonzero(l1, l2, lo):
+                    T = _embed_pair(l1, l2, lo)
+                    Ts = 0.5 * (T + T.transpose(1, 0, 2))
+                    if np.linalg.norm(Ts) > 1e-9:
+                        trees.append((l1, l2))
+                        sym_flat.append(Ts.ravel())
+    else:
+        for l1 in range(L_MAX + 1):
+            for l2 in range(L_MAX + 1):
+                for L in range(abs(l1 - l2), min(l1 + l2, 2 * L_MAX) + 1):
+                    if not cg_nonzero(l1, l2, L):
+                        continue
+                    for l3 in range(L_MAX + 1):
+                        if (l1 + l2 + l3 + lo) % 2 != 0:
+                            continue
+                        if not cg_nonzero(L, l3, lo):
+                            continue
+                        T = _embed_triple(l1, l2, L, l3, lo)
+                        Ts = sum(T.transpose(p + (3,)) for p in
+                                 [(0, 1, 2), (0, 2, 1), (1, 0, 2),
+                                  (1, 2, 0), (2, 0, 1), (2, 1, 0)]) / 6.0
+                        if np.linalg.norm(Ts) > 1e-9:
+                            trees.append((l1, l2, L, l3))
+                            sym_flat.append(Ts.ravel())
+    if not trees:
+        return [], np.zeros((0, 0))
+    S = np.stack(sym_flat, axis=1)                     # [dim, T]
+    Bf = B.reshape(-1, P)                              # [dim, P]
+    M, *_ = np.linalg.lstsq(S, Bf, rcond=None)         # S @ M = Bf
+    resid = np.abs(S @ M - Bf).max()
+    assert resid < 1e-9, f"tree decomposition failed: {resid}"
+    return trees, M.T                                  # [P, T]
+
+
 # ---------------------------------------------------------------------------
 # Wigner D in this basis (tests + weight-free utilities)
 # ---------------------------------------------------------------------------

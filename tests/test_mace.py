@@ -1,0 +1,187 @@
+"""MACE path correctness on CPU.
+
+Parity standard (same as CHGNet — SURVEY §8(c)): mace-torch is not
+installable and the reference ships no numeric tests, so the oracle
+(oracle/mace_ref.py, straight-line dense restatement) is the executable
+definition; the product engine (distmlip_amd/mace_runtime.py, partition
+orchestration + tree-factored contractions) must reproduce it, and
+physics invariants (rotation/translation invariance, equivariant forces,
+Newton's third law) pin both against first principles the restatement
+cannot fake.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from distmlip_amd.mace_model import MACEConfig, MACECore
+from distmlip_amd.structures import diamond_si, random_cell
+from oracle.chgnet_ref import CpuRefOps
+from oracle.graph_ref import brute_force_neighbors
+from oracle.mace_ref import mace_oracle_forward
+
+
+def _small_core(n_elements=3, seed=0, channels=16):
+    cfg = MACEConfig(n_elements=n_elements, channels=channels,
+                     avg_num_neighbors=20.0,
+                     atomic_inter_scale=0.7, atomic_inter_shift=0.1)
+    return MACECore.seeded(cfg, seed=seed).double()
+
+
+def _graph(s):
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 0.0)
+    return g["src"], g["dst"], g["offsets"]
+
+
+def test_oracle_translation_invariance():
+    s = random_cell(40, a=12.0, n_species=3, seed=1)
+    core = _small_core()
+    src, dst, off = _graph(s)
+    r0 = mace_oracle_forward(core, s, src, dst, off)
+    s2 = s
+    shift = np.array([0.13, 0.22, 0.31])
+    s2.frac_coords = np.mod(s.frac_coords + shift, 1.0)
+    # same graph topology (rigid shift) — rebuild to be safe
+    src2, dst2, off2 = _graph(s2)
+    r1 = mace_oracle_forward(core, s2, src2, dst2, off2)
+    assert abs(r0["energy"].item() - r1["energy"].item()) < 1e-9
+
+
+def test_oracle_rotation_invariance_and_force_covariance():
+    from distmlip_amd.so3 import random_rotation
+    from distmlip_amd.structures import Structure
+
+    s = random_cell(40, a=12.0, n_species=3, seed=2)
+    core = _small_core(seed=3)
+    src, dst, off = _graph(s)
+    r0 = mace_oracle_forward(core, s, src, dst, off)
+
+    R = random_rotation(4)
+    lat2 = s.lattice @ R.T
+    s2 = Structure(frac_coords=s.frac_coords.copy(), lattice=lat2,
+                   species=s.species.copy(), pbc=s.pbc.copy())
+    r1 = mace_oracle_forward(core, s2, src, dst, off)
+    # energy invariant, forces covariant: F' = F @ R^T
+    assert abs(r0["energy"].item() - r1["energy"].item()) < 1e-9
+    dF = np.abs(r1["forces"].numpy() - r0["forces"].numpy() @ R.T).max()
+    assert dF < 1e-9, dF
+
+
+def test_oracle_newton_third_law():
+    s = random_cell(40, a=12.0, n_species=3, seed=5)
+    core = _small_core(seed=6)
+    src, dst, off = _graph(s)
+    r = mace_oracle_forward(core, s, src, dst, off)
+    assert np.abs(r["forces"].numpy().sum(0)).max() < 1e-10
+
+
+def test_oracle_forces_vs_finite_difference():
+    s = random_cell(24, a=11.0, n_species=2, seed=7)
+    core = _small_core(n_elements=2, seed=8, channels=8)
+    src, dst, off = _graph(s)
+    r = mace_oracle_forward(core, s, src, dst, off)
+    inv_lat = np.linalg.inv(s.lattice)
+    h = 1e-5
+    rng = np.random.default_rng(0)
+    for atom in rng.choice(s.num_atoms, 3, replace=False):
+        for ax in range(3):
+            sp = s.frac_coords.copy()
+            dm = np.zeros(3)
+            dm[ax] = h
+            sp[atom] += dm @ inv_lat
+            s.frac_coords, keep = sp, s.frac_coords
+            ep = mace_oracle_forward(core, s, src, dst, off,
+                                     compute_forces=False)["energy"].item()
+            sp2 = keep.copy()
+            sp2[atom] -= dm @ inv_lat
+            s.frac_coords = sp2
+            em = mace_oracle_forward(core, s, src, dst, off,
+                                     compute_forces=False)["energy"].item()
+            s.frac_coords = keep
+            fd = -(ep - em) / (2 * h)
+            got = r["forces"][atom, ax].item()
+            assert abs(fd - got) < 1e-5 * max(1.0, abs(got)), (atom, ax)
+
+
+@pytest.mark.parametrize("world", [1, 2])
+def test_mace_engine_vs_oracle_singleproc(world):
+    """Engine (partition orchestration, tree-factored contraction,
+    CpuRefOps backend) at world=1 vs the dense oracle; world=2 runs via
+    the spawn test below."""
+    if world != 1:
+        pytest.skip("world>1 covered by test_mace_spmd_gloo")
+    s = diamond_si((6, 2, 2), jitter=0.1, seed=2)
+    s.species = np.asarray(s.species) % 3
+    core = _small_core(seed=9, channels=16)
+    from distmlip_amd.mace_runtime import MaceSpmdEngine
+    eng = MaceSpmdEngine(core, world=1, threads=2, device="cpu",
+                         ops=CpuRefOps())
+    out = eng.step(s, calc_stresses=True)
+    src, dst, off = _graph(s)
+    ref = mace_oracle_forward(core, s, src, dst, off, compute_stress=True)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
+    F = np.zeros((s.num_atoms, 3))
+    F[out["global_ids_owned"]] = out["forces_owned"].numpy()
+    assert np.abs(F - ref["forces"].numpy()).max() < 1e-10
+    assert np.abs(out["stress"].numpy() - ref["stress"].numpy()).max() < 1e-9
+
+
+def _worker(rank, world, init_file, out_dir):
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from distmlip_amd.mace_runtime import MaceSpmdEngine
+
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        s = diamond_si((12, 2, 2), jitter=0.12, seed=3)
+        s.species = np.asarray(s.species) % 3
+        core = _small_core(seed=9, channels=16)
+        eng = MaceSpmdEngine(core, world, threads=2, device="cpu",
+                             ops=CpuRefOps())
+        out = eng.step(s, calc_stresses=True)
+        np.save(os.path.join(out_dir, f"E_{rank}.npy"),
+                np.array([out["energy"].item()]))
+        np.save(os.path.join(out_dir, f"F_{rank}.npy"),
+                out["forces_owned"].numpy())
+        np.save(os.path.join(out_dir, f"gids_{rank}.npy"),
+                out["global_ids_owned"])
+        np.save(os.path.join(out_dir, f"S_{rank}.npy"),
+                out["stress"].numpy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_mace_spmd_gloo(world, tmp_path):
+    """SPMD MACE: per-rank focused builds + per-layer halo must reproduce
+    the full-graph oracle exactly (fp64) — the reference contract of
+    models.py:135-171 over dist.py:323-358."""
+    init_file = str(tmp_path / "pg_init")
+    mp.spawn(_worker, args=(world, init_file, str(tmp_path)),
+             nprocs=world, join=True)
+
+    s = diamond_si((12, 2, 2), jitter=0.12, seed=3)
+    s.species = np.asarray(s.species) % 3
+    core = _small_core(seed=9, channels=16)
+    src, dst, off = _graph(s)
+    ref = mace_oracle_forward(core, s, src, dst, off, compute_stress=True)
+
+    F = np.zeros((s.num_atoms, 3))
+    covered = np.zeros(s.num_atoms, dtype=bool)
+    for r in range(world):
+        E_r = np.load(f"{tmp_path}/E_{r}.npy")[0]
+        assert abs(E_r - ref["energy"].item()) < 1e-9
+        gids = np.load(f"{tmp_path}/gids_{r}.npy")
+        F[gids] = np.load(f"{tmp_path}/F_{r}.npy")
+        assert not covered[gids].any()
+        covered[gids] = True
+    assert covered.all()
+    dF = np.abs(F - ref["forces"].numpy()).max()
+    assert dF < 1e-10, f"MACE SPMD force error {dF}"
+    S = np.load(f"{tmp_path}/S_0.npy")
+    assert np.abs(S - ref["stress"].numpy()).max() < 1e-9
