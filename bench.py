@@ -80,7 +80,8 @@ def parse_args():
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--workload", type=str, default="si1m",
-                    choices=["li100k", "si1m", "si2m", "si1k"])
+                    choices=["li100k", "si1m", "si2m", "si1k",
+                             "mace62k", "mace500k"])
     ap.add_argument("--threads", type=int, default=0,
                     help="graph-builder threads (0 = cpu_count/world)")
     ap.add_argument("--verlet", type=float, default=0.0, metavar="SKIN",
@@ -95,11 +96,13 @@ def parse_args():
 
 
 class SegSumTimer:
-    """HIP-event timing of the judged scatter-add kernel (D=64 launches)."""
+    """HIP-event timing of the judged scatter-add kernel (D=64 launches
+    for CHGNet; any D for the MACE message scatters)."""
 
-    def __init__(self):
+    def __init__(self, D: int = 64):
         self.events = []     # (start, stop, E_rows, N_rows, D, has_base)
         self.enabled = False
+        self.D = D
 
     def wrap(self):
         import distmlip_amd.ops as ops
@@ -107,7 +110,7 @@ class SegSumTimer:
         timer = self
 
         def timed(msg, row_ptr, n_rows, base=None):
-            if not timer.enabled or msg.shape[-1] != 64:
+            if not timer.enabled or (timer.D and msg.shape[-1] != timer.D):
                 return orig(msg, row_ptr, n_rows, base)
             s = torch.cuda.Event(enable_timing=True)
             e = torch.cuda.Event(enable_timing=True)
@@ -152,7 +155,7 @@ class SegSumTimer:
             "traffic": total_bytes / n * measured_traffic_ratio,
             "traffic_note": "PMC-derived (profiles/r06_pmc_*.csv); gfx950 "
                             "FETCH_SIZE doubled per MI355X_MICROARCH.md",
-            "kernel": "dm_seg_sum_f32(D=64)",
+            "kernel": f"dm_seg_sum_f32(D={self.D or 'any'})",
             "launches": n,
             "avg_ms": total_ms / n,
             "algorithmic_bytes_per_launch": total_bytes / n,
@@ -162,13 +165,37 @@ class SegSumTimer:
 def cpu_baseline_leg(workload_name, threads):
     """Time the ORACLE (CPU restatement, kind='port') on a bounded sample of
     the same workload family and scale to atom-steps/s."""
+    import numpy as _np
+
     from distmlip_amd.dist import Distributed
+    from distmlip_amd.structures import diamond_si
+
+    torch.set_num_threads(threads)
+    if workload_name.startswith("mace"):
+        from distmlip_amd.mace_model import MACECore
+        from oracle.mace_ref import mace_oracle_forward
+
+        s = diamond_si(4, jitter=0.1, seed=0)    # 512-atom bounded sample
+        rng = _np.random.default_rng(77)
+        s.species = rng.integers(0, 3, size=s.num_atoms).astype(_np.int64)
+        t0 = time.time()
+        d = Distributed.create_distributed(
+            s.cart_coords, s.frac_coords, s.lattice, 1, s.pbc, 6.0, 0.0,
+            use_bond_graph=False, num_threads=threads)
+        core = MACECore.seeded(seed=0).float()
+        mace_oracle_forward(core, s, d.py_index_1, d.py_index_2,
+                            d.py_offsets, dtype=torch.float32)
+        dt = time.time() - t0
+        return {"value": s.num_atoms / dt, "unit": "atom_steps_per_s",
+                "cores": threads, "kind": "port",
+                "sample": f"diamond-Si {s.num_atoms} atoms, 1 full MACE "
+                          f"E+F forward (graph build + dense oracle "
+                          f"restatement), {dt:.2f}s"}
+
     from distmlip_amd.model import CHGNetCore
-    from distmlip_amd.structures import diamond_si, workload
     from oracle.chgnet_ref import oracle_forward
 
     s = diamond_si(6, jitter=0.1, seed=0)        # 1,728-atom bounded sample
-    torch.set_num_threads(threads)
     t0 = time.time()
     d = Distributed.create_distributed(
         s.cart_coords, s.frac_coords, s.lattice, 1, s.pbc, 6.0, 3.0,
@@ -198,21 +225,30 @@ def main():
     # host regresses 5x (NUMA + team spawn) — cap unless told otherwise
     threads = args.threads or max(1, min(32, (os.cpu_count() or 8) // max(world, 1)))
 
-    from distmlip_amd.model import CHGNetCore
     from distmlip_amd.structures import workload
 
-    s = workload(args.workload, n_gpus=n_gpus if args.workload == "li100k" else 1)
-    core = CHGNetCore.seeded(seed=0).float()
-    if not use_bg:
-        core.config.use_bond_graph = False
+    is_mace = args.workload.startswith("mace")
+    weak = args.workload in ("li100k", "mace62k")
+    s = workload(args.workload, n_gpus=n_gpus if weak else 1)
 
-    from distmlip_amd.runtime import SpmdEngine
     if world > 1:
         torch.distributed.init_process_group("nccl")
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         torch.cuda.set_device(local_rank)
-    engine = SpmdEngine(core.float(), world, threads=threads,
-                        use_bond_graph=use_bg)
+
+    if is_mace:
+        from distmlip_amd.mace_model import MACECore
+        from distmlip_amd.mace_runtime import MaceSpmdEngine
+        core = MACECore.seeded(seed=0).float()
+        engine = MaceSpmdEngine(core, world, threads=threads)
+    else:
+        from distmlip_amd.model import CHGNetCore
+        from distmlip_amd.runtime import SpmdEngine
+        core = CHGNetCore.seeded(seed=0).float()
+        if not use_bg:
+            core.config.use_bond_graph = False
+        engine = SpmdEngine(core.float(), world, threads=threads,
+                            use_bond_graph=use_bg)
 
     def step():
         return engine.step(s)
@@ -253,7 +289,7 @@ def main():
                       f"{len(v)} calls", file=sys.stderr)
         atexit.register(report)
 
-    timer = SegSumTimer().wrap()
+    timer = SegSumTimer(D=0 if is_mace else 64).wrap()
 
     graph_ms, model_ms = [], []
     for _ in range(args.warmup):
@@ -307,7 +343,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
-            "scaling": "weak" if args.workload == "li100k" else "strong",
+            "scaling": "weak" if weak else "strong",
             "vs_baseline": None,
             "dtype": "f32",
             "data": "synthetic",
@@ -317,9 +353,11 @@ def main():
                 "verlet_rebuilds": (getattr(engine, "_vcache", None) or
                                     {}).get("rebuilds"),
                 "n_atoms": int(total_atoms),
+                "model": "mace-mp-0-medium-shape" if is_mace
+                         else "chgnet-shape",
                 "cutoff": 6.0,
-                "three_body_cutoff": 3.0,
-                "use_bond_graph": use_bg,
+                "three_body_cutoff": None if is_mace else 3.0,
+                "use_bond_graph": False if is_mace else use_bg,
                 "builder_threads": threads,
                 "parallelism": f"graph-parallel slab dp{n_gpus}",
             },

@@ -91,6 +91,23 @@ class Distributed:
 
     # -- construction -----------------------------------------------------
 
+    @staticmethod
+    def cartesian_to_wrapped_fractional(positions_cartesian, lattice, pbc):
+        """Cartesian -> fractional, wrapped on periodic axes (reference
+        dist.py:129-157).  Divergence: the reference short-circuits the
+        fully non-periodic case by returning CARTESIAN coordinates; we
+        always convert properly (the huge padded cell the MACE adapter
+        builds makes the reference's shortcut approximately work — ours
+        is exact for the same inputs)."""
+        lattice = np.asarray(lattice, dtype=float)
+        frac = np.linalg.solve(lattice.T,
+                               np.transpose(positions_cartesian)).T
+        for i, periodic in enumerate(np.asarray(pbc).astype(bool)):
+            if periodic:
+                frac[:, i] %= 1.0
+                frac[:, i] %= 1.0
+        return frac
+
     @classmethod
     def create_distributed(
         cls,

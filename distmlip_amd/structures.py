@@ -128,4 +128,19 @@ def workload(name: str, n_gpus: int = 1, seed: int = 0) -> Structure:
         return diamond_si(50, jitter=0.1, seed=seed)
     if name == "si2m":
         return diamond_si(63, jitter=0.1, seed=seed)   # 2,000,376 atoms
+    if name == "mace500k":
+        # config #4 — MACE-MP-0 medium, 500k atoms fixed (8-GPU strong):
+        # diamond Si 40^3 x 8 = 512,000 atoms, 3 random species
+        s = diamond_si(40, jitter=0.1, seed=seed)
+        rng = np.random.default_rng(seed + 77)
+        s.species = rng.integers(0, 3, size=s.num_atoms).astype(np.int64)
+        return s
+    if name == "mace62k":
+        # per-GPU unit of config #4 (500k/8), weak scaling with n_gpus
+        target = 512000 // 8 * n_gpus
+        reps = int(round((target / 8) ** (1.0 / 3.0)))
+        s = diamond_si(reps, jitter=0.1, seed=seed)
+        rng = np.random.default_rng(seed + 77)
+        s.species = rng.integers(0, 3, size=s.num_atoms).astype(np.int64)
+        return s
     raise ValueError(f"unknown workload {name!r}")

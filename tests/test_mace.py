@@ -129,6 +129,40 @@ def test_mace_engine_vs_oracle_singleproc(world):
     assert np.abs(out["stress"].numpy() - ref["stress"].numpy()).max() < 1e-9
 
 
+@pytest.mark.parametrize("P", [1, 2])
+def test_mace_dist_api_mirror_vs_oracle(P):
+    """The reference-shaped plugin surface (MACE_Dist.from_existing /
+    enable_distributed_mode / dist_forward, models.py:40-263 mirror;
+    get_neighborhood_dist, mace_utils.py:25-78 mirror) against the
+    oracle on a periodic cell, P cpu partitions in one process."""
+    from distmlip_amd.mace import MACE_Dist, get_neighborhood_dist
+
+    s = diamond_si((8, 2, 2), jitter=0.1, seed=4)
+    s.species = np.asarray(s.species) % 3
+    core = _small_core(seed=11, channels=16)
+
+    pos_np = s.frac_coords @ s.lattice
+    edge_index, shifts, unit_shifts, cell, dist_info = \
+        get_neighborhood_dist(pos_np, 6.0, pbc=(True, True, True),
+                              cell=s.lattice.copy(), num_partitions=P)
+    model = MACE_Dist.from_existing(core)
+    model.enable_distributed_mode(["cpu"] * P)
+
+    positions = torch.tensor(pos_np, dtype=torch.float64,
+                             requires_grad=True)
+    data = {"positions": positions,
+            "species": torch.tensor(np.asarray(s.species),
+                                    dtype=torch.long),
+            "shifts": torch.tensor(shifts, dtype=torch.float64)}
+    out = model.dist_forward(data, dist_info)
+
+    ref = mace_oracle_forward(core, s, dist_info.py_index_1,
+                              dist_info.py_index_2, dist_info.py_offsets)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
+    dF = (out["forces"] - ref["forces"]).abs().max().item()
+    assert dF < 1e-10, dF
+
+
 def _worker(rank, world, init_file, out_dir):
     import sys
     sys.path.insert(0, os.path.dirname(os.path.dirname(
