@@ -77,12 +77,35 @@ class MaceSpmdEngine:
         ft = self.float_th
         C = cfg.channels
 
+        gpu_pd = None
         if dist_info is None:
-            dist_info = self.build_graph(structure)
-        pd = PartitionData(dist_info, r, dev, use_bond_graph=False)
-        plan = halo_plan(dist_info.markers[r], r, P)
-        gids = np.asarray(dist_info.global_ids[r])
-        n_owned = dist_info.num_owned_atoms(r)
+            from distmlip_amd import gpu_graph
+            if (dev.type == "cuda"
+                    and not getattr(self.ops, "is_reference", False)
+                    and gpu_graph.supported(structure, cfg.r_max)):
+                if P == 1:
+                    gpu_pd = gpu_graph.build(structure, cfg.r_max, 0.0,
+                                             1e-8, False, dev)
+                else:
+                    gpu_pd = gpu_graph.build_partition(
+                        structure, P, r, cfg.r_max, 0.0, 1e-8, False, dev)
+            else:
+                dist_info = self.build_graph(structure)
+        if gpu_pd is not None and P == 1:
+            pd = gpu_pd
+            plan = []
+            gids = np.arange(pd.n_atoms)
+            n_owned = pd.n_atoms
+        elif gpu_pd is not None:
+            pd = gpu_pd
+            plan = halo_plan(pd.markers, r, P)
+            gids = pd.global_ids
+            n_owned = pd.n_owned
+        else:
+            pd = PartitionData(dist_info, r, dev, use_bond_graph=False)
+            plan = halo_plan(dist_info.markers[r], r, P)
+            gids = np.asarray(dist_info.global_ids[r])
+            n_owned = dist_info.num_owned_atoms(r)
         halo_seq = _HaloSeq()
 
         def _halo_dict(x: Dict[int, torch.Tensor]) -> Dict[int, torch.Tensor]:
@@ -115,9 +138,10 @@ class MaceSpmdEngine:
         spec = np.asarray(structure.species)
         species = torch.tensor(spec[gids], dtype=torch.long, device=dev)
 
-        csr = dist_info.csr_parts[r] if getattr(dist_info, "csr_parts",
-                                                None) else None
-        if csr is not None:
+        if gpu_pd is not None:
+            off_local = pd.off_i8.to(ft)
+        elif (csr := dist_info.csr_parts[r]
+              if getattr(dist_info, "csr_parts", None) else None) is not None:
             off_local = torch.from_numpy(csr["offsets_i8"]).to(dev).to(ft)
         else:
             egids = np.asarray(dist_info.L2G_DE_mapping_list[r])

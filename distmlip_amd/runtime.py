@@ -299,21 +299,38 @@ class SpmdEngine:
         gpu_pd = None
         if dist_info is None:
             from distmlip_amd import gpu_graph
-            if (P == 1 and self.gpu_build == "auto"
+            if (self.gpu_build == "auto"
                     and self.graph_backend is None
                     and dev.type == "cuda"
                     and gpu_graph.supported(structure, cfg.cutoff)):
-                gpu_pd = gpu_graph.build(
-                    structure, cfg.cutoff, cfg.three_body_cutoff, 1e-8,
-                    self.use_bond_graph, dev, frac_override=frac_override)
+                if P == 1:
+                    gpu_pd = gpu_graph.build(
+                        structure, cfg.cutoff, cfg.three_body_cutoff, 1e-8,
+                        self.use_bond_graph, dev,
+                        frac_override=frac_override)
+                else:
+                    # GPU SPMD slab build: full-box GPU NL + on-device
+                    # partition assembly (replaces the 0.4 s/rank CPU
+                    # focused build — the N=8 scaling bottleneck)
+                    gpu_pd = gpu_graph.build_partition(
+                        structure, P, r, cfg.cutoff,
+                        cfg.three_body_cutoff, 1e-8, self.use_bond_graph,
+                        dev, frac_override=frac_override)
             else:
                 dist_info = self.build_graph(structure)
         halo_seq = _HaloSeq()                 # fresh guard per step
-        if gpu_pd is not None:
+        if gpu_pd is not None and P == 1:
             pd = gpu_pd
             plan, line_plan = [], []
             gids = None                       # identity: local ids == global
             n_owned = pd.n_atoms
+        elif gpu_pd is not None:
+            pd = gpu_pd
+            plan = halo_plan(pd.markers, r, P)
+            line_plan = halo_plan(pd.line_markers, r, P) \
+                if self.use_bond_graph else None
+            gids = pd.global_ids
+            n_owned = pd.n_owned
         else:
             pd = PartitionData(dist_info, r, dev, self.use_bond_graph)
             plan = halo_plan(dist_info.markers[r], r, P)
