@@ -65,6 +65,18 @@ def _install_stderr_filter(patterns):
 
     threading.Thread(target=pump, daemon=True).start()
 
+    def restore():
+        # put the real stderr back before interpreter teardown: tools
+        # that print at process finalization (rocprofv3 stats, C library
+        # destructors) must not write into the dying pipe
+        import time
+        time.sleep(0.05)          # let the pump drain in-flight lines
+        os.dup2(real, 2)
+        os.close(r)
+
+    import atexit
+    atexit.register(restore)
+
 
 _install_stderr_filter([b"matrix and stride size must be positive"])
 
