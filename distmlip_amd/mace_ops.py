@@ -53,6 +53,14 @@ def cg_t(l1, l2, l3, like: torch.Tensor) -> torch.Tensor:
                  str(like.device))
 
 
+def _chan_mix(xl: torch.Tensor, w: torch.Tensor, s: float) -> torch.Tensor:
+    """out[n,m,d] = s * sum_c w[m,c] x[n,c,d] as one tall GEMM
+    ([N*d, C] x [C, C]) — never a batched tiny-tile einsum."""
+    N, C, d = xl.shape
+    y = xl.transpose(1, 2).reshape(N * d, C) @ w.t()
+    return (y * s).view(N, d, C).transpose(1, 2)
+
+
 def irreps_linear(lin, x: Dict[int, torch.Tensor], norm: bool = True
                   ) -> Dict[int, torch.Tensor]:
     """o3.Linear: per-l channel mix, 1/sqrt(C) path normalization."""
@@ -60,7 +68,7 @@ def irreps_linear(lin, x: Dict[int, torch.Tensor], norm: bool = True
     for l in lin.ls:
         w = lin.w[str(l)]
         s = 1.0 / math.sqrt(w.shape[1]) if norm else 1.0
-        out[l] = torch.einsum("ncd,mc->nmd", x[l], w) * s
+        out[l] = _chan_mix(x[l], w, s)
     return out
 
 
@@ -75,10 +83,11 @@ def skip_tp(inter: Interaction, x: Dict[int, torch.Tensor],
     for l in inter.skip_ls:
         W = inter.skip[str(l)]                       # [n_elem, C, C]
         xl = x[l]
-        ol = torch.empty_like(xl)
+        ol = torch.zeros_like(xl)
         for e in torch.unique(species):
             m = species == e
-            ol[m] = torch.einsum("ncd,mc->nmd", xl[m], W[e]) * s
+            ol = ol.index_put((m.nonzero(as_tuple=True)[0],),
+                              _chan_mix(xl[m], W[e], s))
         out[l] = ol
     return out
 
@@ -93,15 +102,27 @@ def conv_tp_messages(inter: Interaction, gathered: Dict[int, torch.Tensor],
     gathers via the ops backend so the backward is a segmented
     reduction); Y: [E,16] SH of edge vectors; tp_w: [E, P, C] per-edge
     path weights (radial MLP output).  Returns per-l3 messages
-    [E, C, d3]."""
+    [E, C, d3].
+
+    Shape discipline (measured, profiles/r2_mace_kernel_stats.csv): a
+    batch-E bmm of (C x d1)(d1 x d3) tiles runs rocBLAS at ~1% of peak
+    (MT16x16 strided-batched tiles, ~100 ms per call at E=2.9M) — 96% of
+    a 46 s step.  yc stays a tall-skinny GEMM ([E,d2] x [d2, d1*d3]);
+    the k1-contraction (d1 <= 3) is unrolled into broadcast
+    multiply-accumulates, which are plain HBM-bound elementwise
+    kernels."""
     msgs: Dict[int, torch.Tensor] = {}
     for p, (l1, l2, l3) in enumerate(inter.paths):
         CG = cg_t(l1, l2, l3, Y)
         o2, d2 = so3.L_OFF[l2], so3.L_DIMS[l2]
+        d1, d3 = so3.L_DIMS[l1], so3.L_DIMS[l3]
         yb = Y[:, o2:o2 + d2]
-        yc = torch.einsum("ef,dfg->edg", yb, CG)     # [E, d1, d3]
-        xa = gathered[l1] * tp_w[:, p, :].unsqueeze(-1)
-        m = torch.bmm(xa, yc)                        # [E, C, d3]
+        # [E, d2] @ [d2, d1*d3] -> [E, d1, d3]
+        yc = (yb @ CG.permute(1, 0, 2).reshape(d2, d1 * d3)).view(-1, d1, d3)
+        xa = gathered[l1] * tp_w[:, p, :].unsqueeze(-1)    # [E, C, d1]
+        m = xa[:, :, 0:1] * yc[:, 0, :].unsqueeze(1)
+        for k1 in range(1, d1):
+            m = m + xa[:, :, k1:k1 + 1] * yc[:, k1, :].unsqueeze(1)
         if l3 in msgs:
             msgs[l3] = msgs[l3] + m
         else:
@@ -116,15 +137,57 @@ def _trees_M(nu: int, lo: int, dtype_str: str, device_str: str):
                                device=torch.device(device_str))
 
 
+@lru_cache(maxsize=None)
+def _combo_groups(nu: int, lo: int, dtype_str: str, device_str: str):
+    """Trees grouped by their (l1[,l2[,l3]]) slot signature, each group's
+    coupling tensors stacked into ONE [d1*d2*d3, n_t*do] GEMM operand.
+    Returns list of (combo, tree_indices, D) with D on device."""
+    trees, _ = so3.symmetric_basis_trees(nu, lo)
+    do = 2 * lo + 1
+    dt = getattr(torch, dtype_str)
+    dev = torch.device(device_str)
+    groups = {}
+    for t, tr in enumerate(trees):
+        combo = tr if nu == 1 else (tr[:2] if nu == 2
+                                    else (tr[0], tr[1], tr[3]))
+        groups.setdefault(combo, []).append(t)
+    out = []
+    for combo, tidx in groups.items():
+        Ds = []
+        for t in tidx:
+            tr = trees[t]
+            if nu == 1:
+                D = np.eye(2 * tr[0] + 1)
+            elif nu == 2:
+                D = so3.real_cg(tr[0], tr[1], lo).reshape(-1, do)
+            else:
+                l1, l2, L, l3 = tr
+                C12 = so3.real_cg(l1, l2, L)
+                C3 = so3.real_cg(L, l3, lo)
+                D = np.einsum("abL,Lko->abko", C12, C3).reshape(-1, do)
+            Ds.append(D)
+        D = np.concatenate(Ds, axis=1)               # [prod d, n_t*do]
+        out.append((combo, tidx,
+                    torch.tensor(D, dtype=dt, device=dev)))
+    return out
+
+
 def symmetric_contract(prod: ProductBasis, x: Dict[int, torch.Tensor],
                        species: torch.Tensor, correlation: int
                        ) -> Dict[int, torch.Tensor]:
     """MACE SymmetricContraction: sum over nu=1..corr of per-element
-    weighted symmetric couplings, evaluated through the tree
-    factorization (so3.symmetric_basis_trees) — no dense [16]^nu tensors
-    at run time."""
+    weighted symmetric couplings via the tree factorization
+    (so3.symmetric_basis_trees) — no dense [16]^nu tensors at run time.
+
+    Shape discipline: per slot-signature group, the couplings collapse to
+    one tall-skinny GEMM [N*C, d1*d2*d3] x [d1*d2*d3, n_t*do] over the
+    broadcast outer product of the slot blocks; per-tree weighting is a
+    broadcast multiply.  (Per-tree einsums lowered to batch-N bmms of
+    tiny tiles — the rocBLAS ~1%-of-peak shape, see
+    conv_tp_messages.)"""
     like = next(iter(x.values()))
     N, C = like.shape[0], like.shape[1]
+    dts, devs = str(like.dtype).split(".")[-1], str(like.device)
     out = {}
     for lo in prod.out_ls:
         do = 2 * lo + 1
@@ -133,28 +196,30 @@ def symmetric_contract(prod: ProductBasis, x: Dict[int, torch.Tensor],
             key = f"{lo}_{nu}"
             if key not in prod.weights:
                 continue
-            trees, M = _trees_M(nu, lo, str(like.dtype).split(".")[-1],
-                                str(like.device))
-            tvs = []
-            for tr in trees:
+            trees, M = _trees_M(nu, lo, dts, devs)
+            T = len(trees)
+            tvs: list = [None] * T
+            for combo, tidx, D in _combo_groups(nu, lo, dts, devs):
                 if nu == 1:
-                    tvs.append(x[tr[0]])
+                    xx = x[combo[0]]
                 elif nu == 2:
-                    l1, l2 = tr
-                    CG = cg_t(l1, l2, lo, like)
-                    tvs.append(torch.einsum("abo,nca,ncb->nco",
-                                            CG, x[l1], x[l2]))
+                    l1, l2 = combo
+                    xx = (x[l1].unsqueeze(-1) * x[l2].unsqueeze(-2)
+                          ).reshape(N, C, -1)
                 else:
-                    l1, l2, L, l3 = tr
-                    C12 = cg_t(l1, l2, L, like)
-                    C3 = cg_t(L, l3, lo, like)
-                    t2 = torch.einsum("abL,nca,ncb->ncL", C12, x[l1], x[l2])
-                    t3 = torch.einsum("Lko,nck->ncLo", C3, x[l3])
-                    tvs.append(torch.einsum("ncL,ncLo->nco", t2, t3))
-            tv = torch.stack(tvs, dim=-1)            # [N, C, do, T]
+                    l1, l2, l3 = combo
+                    xx = (x[l1].unsqueeze(-1) * x[l2].unsqueeze(-2)
+                          ).reshape(N, C, -1)
+                    xx = (xx.unsqueeze(-1) * x[l3].unsqueeze(-2)
+                          ).reshape(N, C, -1)
+                g = xx @ D                            # [N, C, n_t*do]
+                for j, t in enumerate(tidx):
+                    tvs[t] = g[:, :, j * do:(j + 1) * do]
             w = prod.weights[key]                    # [n_elem, P, C]
             wM = torch.einsum("epc,pt->etc", w, M)   # [n_elem, T, C]
-            acc = acc + torch.einsum("ncot,ntc->nco", tv, wM[species])
+            wMn = wM[species]                        # [N, T, C]
+            tv = torch.stack(tvs, dim=-1)            # [N, C, do, T]
+            acc = acc + (tv * wMn.permute(0, 2, 1).unsqueeze(2)).sum(-1)
         out[lo] = acc
     return out
 
