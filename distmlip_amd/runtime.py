@@ -499,6 +499,40 @@ class SpmdEngine:
                        torch.tensor(layer_i % cfg.n_blocks), v, e,
                        bond_expansion)
 
+        # -- whole-graph hand-sequenced reverse (single partition, no
+        # verlet masks): every block + the final conv in ONE Function
+        # whose backward is sequenced by hand (conv.py)
+        from distmlip_amd.conv import _WholeGraphFn, whole_graph_available
+        if (P == 1 and not mask_cutoffs and d % 2 == 0
+                and whole_graph_available(
+                    ops, _packed_weights(core.atom_convs[0].edge_mlp))):
+            wbb_m = w_bb if w_bb is not None else \
+                core.bond_bond_weights(bond_expansion)
+            wab_m = w_ab if w_ab is not None else \
+                core.atom_bond_weights(bond_expansion)
+            w3_m = None
+            if use_bg:
+                w3_m = w_3b if w_3b is not None else \
+                    core.threebody_bond_weights(exp3)
+            packs = {"atom": [(_packed_weights(b.edge_mlp),
+                               _packed_weights(b.node_mlp))
+                              for b in core.atom_convs]}
+            if use_bg:
+                packs["bond"] = [_packed_weights(b.bond_mlp)
+                                 for b in core.bond_convs]
+                packs["angle"] = [_packed_weights(b.angle_mlp)
+                                  for b in core.bond_convs]
+            v_final, v_mid = _WholeGraphFn.apply(
+                e, a if use_bg else None, wbb_m.contiguous(),
+                wab_m.contiguous(),
+                w3_m.contiguous() if w3_m is not None else None, v, pd,
+                ops, packs, d, cfg.n_blocks, use_bg, ckpt)
+            site_props = core.sitewise_readout(v_mid)
+            v = v_final
+            return self._finish(structure, core, pos, strain, plan,
+                                species_local, n_owned, gids, v,
+                                site_props, calc_stresses, P)
+
         for layer_i in range(cfg.n_blocks - 1):           # chgnet.py:296-368
             v, e = atom_conv(layer_i, v, e)
             if use_bg:
@@ -555,6 +589,15 @@ class SpmdEngine:
         v, e = atom_conv(-1, v, e)                         # final atom block
         v = _halo(v, plan)
 
+        return self._finish(structure, core, pos, strain, plan,
+                            species_local, n_owned, gids, v, site_props,
+                            calc_stresses, P)
+
+    def _finish(self, structure, core, pos, strain, plan, species_local,
+                n_owned, gids, v, site_props, calc_stresses, P):
+        """Readout + force assembly shared by the op-by-op and
+        whole-graph paths (reference pes.py:101-145 semantics)."""
+        dev = self.device
         atom_e = core.final_layer(v)
         e_local_raw = atom_e[:n_owned].sum()
 
