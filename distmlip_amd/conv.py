@@ -417,12 +417,17 @@ class _WholeGraphFn(torch.autograd.Function):
             v_mid, e_mid = ctx.final_in
             pe, pn = packs["atom"][nb - 1]
             sv_f = ctx.sv_f
-            if sv_f is None:
-                _, _, sv_f = atom_fwd(ops, pd, v_mid, e_mid, wbb, wab,
-                                      pe[:4], pn[:4], d)
-            gv, ge, _, _ = atom_bwd(ops, pd, sv_f, pe[:4], pn[:4], d,
-                                    wbb, wab, go_v2, None,
-                                    acc_wbb=gwbb, acc_wab=gwab)
+            if sv_f is not None:
+                gv, ge, _, _ = atom_bwd(ops, pd, sv_f, pe[:4], pn[:4], d,
+                                        wbb, wab, go_v2, None,
+                                        acc_wbb=gwbb, acc_wab=gwab)
+                ctx.sv_f = None
+            else:
+                # streamed recompute: at most one MLP's [E,2d] saves live
+                gv, ge = atom_bwd_recompute(ops, pd, v_mid, e_mid, wbb,
+                                            wab, pe, pn, d, go_v2, None,
+                                            gwbb, gwab)
+            ctx.final_in = None
             ga = None
             gn = None
 
@@ -431,52 +436,75 @@ class _WholeGraphFn(torch.autograd.Function):
                 sv = ctx.saves[i]
                 pe, pn = packs["atom"][i]
                 if recompute:
-                    # re-run this block's forward, keeping z/cg
-                    v_a, e_a, sv_at = atom_fwd(ops, pd, v_in, e_in, wbb,
-                                               wab, pe[:4], pn[:4], d)
+                    # re-run the pieces of this block's forward the
+                    # reverse needs, lean where possible
                     if use_bg:
+                        v_a, e_a = _lean_atom(ops, pd, v_in, e_in, wbb,
+                                              wab, pe, pn, d)
                         n_c = _edge_to_bond(pd, n_in, e_a)
-                        n_b, sv_b = bond_fwd(ops, pd, n_c, a_in, v_a, w3,
-                                             packs["bond"][i][:4], d)
-                        if i < nb - 2:
-                            _, sv_a = angle_fwd(ops, pd, n_b, a_in, v_a,
-                                                packs["angle"][i][:4], d)
+                        del e_a
+                    sv_at = sv_b = sv_a = None
                 else:
                     sv_at = sv["atom"]
                     sv_b = sv.get("bond")
                     sv_a = sv.get("angle")
+                ctx.block_inputs[i] = None
+                ctx.saves[i] = None
 
                 if use_bg:
                     if ga is None:
-                        ga = torch.zeros_like(ctx.block_inputs[0][3])
+                        ga = torch.zeros(pd.n_lines if hasattr(
+                            pd, "n_lines") else len(pd.l_src), d,
+                            dtype=gv.dtype, device=gv.device)
                     gn_b = gn if gn is not None else \
                         torch.zeros(pd.n_bonds, d, dtype=gv.dtype,
                                     device=gv.device)
                     if i < nb - 2:
+                        if sv_a is None:
+                            n_b = _lean_bond(ops, pd, n_c, a_in, v_a, w3,
+                                             packs["bond"][i], d)
+                            _, sv_a = angle_fwd(ops, pd, n_b, a_in, v_a,
+                                                packs["angle"][i][:4], d)
+                            del n_b
                         # angle reverse: n/v contributions accumulate,
                         # da replaces ga (includes its pass-through)
                         _, ga_new, _ = angle_bwd(
                             ops, pd, sv_a, packs["angle"][i][:4], d, ga,
                             acc_n=gn_b, acc_v=gv)
                         ga = ga_new
+                        sv_a = None
                     # bond_to_edge reverse: e_b = e_a.index_copy(map_de,
                     # n_b[map_ude])
                     gn_b.index_add_(0, pd.map_ude, ge[pd.map_de])
                     ge = ge.index_fill(0, pd.map_de, 0)
                     # bond conv reverse (a/v/w3 contributions accumulate)
+                    if sv_b is None:
+                        _, sv_b = bond_fwd(ops, pd, n_c, a_in, v_a, w3,
+                                           packs["bond"][i][:4], d)
+                        del n_c
                     gn_c, _, _, _ = bond_bwd(ops, pd, sv_b,
                                              packs["bond"][i][:4], d, gn_b,
                                              acc_w3=gw3, acc_a=ga,
                                              acc_v=gv)
+                    sv_b = None
+                    del gn_b
                     # edge_to_bond reverse: n_c = n_in.index_copy(map_ude,
                     # e_a[map_de])
                     ge.index_add_(0, pd.map_de, gn_c[pd.map_ude])
                     gn = gn_c.index_fill(0, pd.map_ude, 0) \
                         if pd.n_bonds != len(pd.map_ude) else None
+                    del gn_c
 
-                gv, ge, _, _ = atom_bwd(ops, pd, sv_at, pe[:4], pn[:4], d,
-                                        wbb, wab, gv, ge,
-                                        acc_wbb=gwbb, acc_wab=gwab)
+                if sv_at is not None:
+                    gv, ge, _, _ = atom_bwd(ops, pd, sv_at, pe[:4], pn[:4],
+                                            d, wbb, wab, gv, ge,
+                                            acc_wbb=gwbb, acc_wab=gwab)
+                else:
+                    if use_bg:
+                        del v_a
+                    gv, ge = atom_bwd_recompute(ops, pd, v_in, e_in, wbb,
+                                                wab, pe, pn, d, gv, ge,
+                                                gwbb, gwab)
 
             # initial n0 = empty.index_copy(map_ude, e0[map_de])
             if use_bg and gn is not None:
@@ -494,3 +522,72 @@ def whole_graph_available(ops, mlp_pack) -> bool:
             and hasattr(ops, "r_gather_add3")
             and not mlp_pack[0].requires_grad
             and torch.is_grad_enabled())
+
+
+def _lean_edge_update(ops, pd, v, e, wbb, pe, d):
+    """e2 only (no z/cg kept) — first half of the atom conv."""
+    wcg1, bcg1, w21, b21, fusedT1 = pe
+    ws1, wd1 = wcg1[:, :d], wcg1[:, d:2 * d]
+    if d == 64 and hasattr(ops, "edge_mlp3_act"):
+        h1 = ops.edge_mlp3_act(e, fusedT1, bcg1, v @ ws1.t(), v @ wd1.t(),
+                               pd)
+    else:
+        h1 = ops.gather_add3_act(v @ ws1.t(), v @ wd1.t(),
+                                 torch.addmm(bcg1, e, wcg1[:, 2 * d:].t()),
+                                 pd)
+    return ops.gated_combine_packed(_second_fwd(h1, w21, b21, d), wbb, e)
+
+
+def atom_bwd_recompute(ops, pd, v_in, e_in, wbb, wab, pe, pn, d,
+                       go_v2, go_e2, acc_wbb, acc_wab):
+    """Streamed recompute + reverse of one atom conv from its INPUTS:
+    the node MLP's z/cg are recomputed, reversed and FREED before the
+    edge MLP's are materialized, so at most one MLP's [E,2d] saves are
+    live — the memory shape that fits si1m's 46M edges (a single
+    all-at-once recompute OOMs at ~94 GB of z/cg)."""
+    ws1, wd1, we1, bcg1, w21, _ = _unpack_edge(pe[:4], d)
+    ws2, wd2, we2, bcg2, w22, _ = _unpack_edge(pn[:4], d)
+    go_v2 = go_v2.contiguous() if go_v2 is not None else None
+    go_e2 = go_e2.contiguous() if go_e2 is not None else None
+
+    if go_v2 is not None:
+        e2 = _lean_edge_update(ops, pd, v_in, e_in, wbb, pe, d)
+        z2, h2 = ops.r_gather_add3(v_in @ ws2.t(), v_in @ wd2.t(),
+                                   torch.addmm(bcg2, e2, we2.t()), pd)
+        del e2
+        cg2 = _second_fwd(h2, w22, pn[3], d)
+        del h2
+        dmsg = ops.r_gather_dst(go_v2, pd)
+        dcg2, dwab = ops.r_combine_bwd(dmsg, cg2, wab)
+        del dmsg, cg2
+        dz2 = ops.r_silu_bwd(_second_bwd(dcg2, w22, d), z2)
+        del dcg2, z2
+        ge2 = (torch.addmm(go_e2, dz2, we2) if go_e2 is not None
+               else dz2 @ we2)
+        acc_wab.add_(dwab)
+        del dwab
+    else:
+        dz2 = None
+        ge2 = go_e2
+
+    z1, h1 = ops.r_gather_add3(v_in @ ws1.t(), v_in @ wd1.t(),
+                               torch.addmm(bcg1, e_in, we1.t()), pd)
+    cg1 = _second_fwd(h1, w21, pe[3], d)
+    del h1
+    dcg1, dwbb = ops.r_combine_bwd(ge2, cg1, wbb)
+    del cg1
+    dz1 = ops.r_silu_bwd(_second_bwd(dcg1, w21, d), z1)
+    del dcg1, z1
+    ge = torch.addmm(ge2, dz1, we1)
+    del ge2
+    gv = torch.addmm(go_v2, ops.r_seg_src(dz1, pd), ws1) \
+        if go_v2 is not None else ops.r_seg_src(dz1, pd) @ ws1
+    gv.addmm_(ops.r_seg_dst(dz1, pd), wd1)
+    del dz1
+    if dz2 is not None:
+        gv.addmm_(ops.r_seg_src(dz2, pd), ws2)
+        gv.addmm_(ops.r_seg_dst(dz2, pd), wd2)
+        del dz2
+    acc_wbb.add_(dwbb)
+    del dwbb
+    return gv, ge

@@ -354,3 +354,78 @@ def random_rotation(seed: int = 0) -> np.ndarray:
     if np.linalg.det(Q) < 0:
         Q[:, 0] = -Q[:, 0]
     return Q
+
+
+# ---------------------------------------------------------------------------
+# batched Wigner D from rotation matrices (round 2, UMA/eSCN path)
+# ---------------------------------------------------------------------------
+
+@lru_cache(maxsize=None)
+def wigner_D_poly(l: int) -> np.ndarray:
+    """T [9^l, dl*dl] with D_l(R).flatten() == kron(R,..,R).flatten() @ T
+    for any ROTATION R: D_l entries are degree-l polynomials in R's
+    entries, so T is solved once (lstsq over sampled rotations against
+    wigner_D_from_sh).  Any solution agreeing on SO(3) has identical
+    tangential derivatives along the manifold, so autograd gradients
+    through batched evaluation are exact."""
+    d = 2 * l + 1
+    K = max(60, 3 * 9 ** l)
+    rng = np.random.default_rng(100 + l)
+    F = np.empty((K, 9 ** l))
+    Y = np.empty((K, d * d))
+    for k in range(K):
+        A = rng.normal(size=(3, 3))
+        Q, r = np.linalg.qr(A)
+        Q *= np.sign(np.diag(r))
+        if np.linalg.det(Q) < 0:
+            Q[:, 0] = -Q[:, 0]
+        f = np.array([1.0])
+        for _ in range(l):
+            f = np.kron(f, Q.ravel())
+        F[k] = f
+        D = wigner_D_from_sh(Q)
+        o = L_OFF[l]
+        Y[k] = D[o:o + d, o:o + d].ravel()
+    T, *_ = np.linalg.lstsq(F, Y, rcond=None)
+    resid = np.abs(F @ T - Y).max()
+    assert resid < 1e-8, f"wigner_D_poly l={l}: residual {resid}"
+    return T
+
+
+def wigner_D_batch(R: torch.Tensor, l_max: int = L_MAX) -> torch.Tensor:
+    """[..., 3, 3] rotations -> [..., S, S] block-diagonal Wigner D over
+    l = 0..l_max (S = sum of 2l+1).  Differentiable in R; evaluation is
+    kron-powers of R times the precomputed coefficient tensors."""
+    S = sum(2 * ll + 1 for ll in range(l_max + 1))
+    batch = R.shape[:-2]
+    Rf = R.reshape(*batch, 9)
+    D = R.new_zeros(*batch, S, S)
+    D[..., 0, 0] = 1.0
+    f = Rf
+    off = 1
+    for l in range(1, l_max + 1):
+        d = 2 * l + 1
+        T = torch.as_tensor(wigner_D_poly(l), dtype=R.dtype,
+                            device=R.device)
+        blk = (f @ T).reshape(*batch, d, d)
+        D[..., off:off + d, off:off + d] = blk
+        off += d
+        if l < l_max:
+            # next kron power: f_{l+1}[..., i*9+j] = f_l[..., i] * Rf[..., j]
+            f = (f.unsqueeze(-1) * Rf.unsqueeze(-2)).reshape(*batch, -1)
+    return D
+
+
+def edge_align_rotation(vec: torch.Tensor) -> torch.Tensor:
+    """[..., 3] -> [..., 3, 3] rotation R with R @ v_hat = z_hat,
+    differentiable a.e. (the reference-axis pick is piecewise constant,
+    like eSCN's perpendicular construction)."""
+    v = vec / torch.linalg.norm(vec, dim=-1, keepdim=True)
+    # reference axis: the canonical axis least aligned with v
+    a = torch.zeros_like(v)
+    idx = v.abs().argmin(dim=-1, keepdim=True)
+    a.scatter_(-1, idx, 1.0)
+    r1 = torch.cross(a, v, dim=-1)
+    r1 = r1 / torch.linalg.norm(r1, dim=-1, keepdim=True)
+    r2 = torch.cross(v, r1, dim=-1)                  # right-handed: r1 x r2 = v
+    return torch.stack([r1, r2, v], dim=-2)
