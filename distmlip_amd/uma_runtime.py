@@ -1,0 +1,221 @@
+"""UMA/eSCN SPMD runtime — one process per GPU over torch.distributed.
+
+MI355X-native replacement for the reference's thread-pool-per-partition
+loop (implementations/uma/escn_md.py:442-500: per-layer block forward +
+atom_transfer halo, ThreadPoolExecutor + CUDA events): each rank owns one
+slab partition; the halo is the same HaloExchange autograd Function the
+CHGNet/MACE engines use, issued once per layer at the reference's
+transfer point (escn_md.py:496 — and after the edge-degree embedding,
+escn_md.py:416).  bf16 runs under autocast (GEMM-heavy ops in bf16,
+reductions in fp32) — the BASELINE config #5 dtype.
+"""
+from __future__ import annotations
+
+from copy import deepcopy
+from typing import Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from distmlip_amd import so3, uma_ops
+from distmlip_amd.chgnet import PartitionData
+from distmlip_amd.dist import Distributed
+from distmlip_amd.ops_base import default_ops_factory
+from distmlip_amd.runtime import HaloExchange, _HaloSeq, _exchange, halo_plan
+from distmlip_amd.uma_model import UMACore
+
+
+def _flat(x):
+    return x.reshape(x.shape[0], -1)
+
+
+class UmaSpmdEngine:
+    """Per-rank UMA E+F engine (config #5 path)."""
+
+    def __init__(self, core: UMACore, world: int, threads: int = 8,
+                 device: Optional[str] = None, ops=None,
+                 checkpoint: str = "auto", autocast_bf16: bool = False):
+        self.rank = dist.get_rank() if dist.is_initialized() else 0
+        assert world == 1 or dist.is_initialized()
+        self.world = world
+        self.config = core.config
+        if device is None:
+            device = f"cuda:{torch.cuda.current_device()}"
+        self.device = torch.device(device)
+        self.core = deepcopy(core).to(self.device).eval()
+        self.core.requires_grad_(False)
+        self.ops = ops if ops is not None else default_ops_factory(self.device)
+        self.threads = threads
+        self.checkpoint = checkpoint
+        self.autocast_bf16 = autocast_bf16
+        self.float_th = self.core.sphere_embedding.dtype
+
+    def build_graph(self, structure) -> Distributed:
+        focus = self.rank if self.world > 1 else -1
+        return Distributed.create_distributed(
+            cart_coords=structure.cart_coords,
+            frac_coords=structure.frac_coords,
+            lattice_matrix=structure.lattice,
+            num_partitions=self.world, pbc=structure.pbc,
+            cutoff=self.config.cutoff, three_body_cutoff=0.0,
+            use_bond_graph=False, num_threads=self.threads,
+            focus_partition=focus)
+
+    def step(self, structure, dist_info: Optional[Distributed] = None,
+             calc_stresses: bool = False):
+        r, P = self.rank, self.world
+        cfg, core, ops, dev = self.config, self.core, self.ops, self.device
+        ft = self.float_th
+        C, S = cfg.sphere_channels, cfg.S
+
+        gpu_pd = None
+        if dist_info is None:
+            from distmlip_amd import gpu_graph
+            if (dev.type == "cuda"
+                    and not getattr(self.ops, "is_reference", False)
+                    and gpu_graph.supported(structure, cfg.cutoff)):
+                if P == 1:
+                    gpu_pd = gpu_graph.build(structure, cfg.cutoff, 0.0,
+                                             1e-8, False, dev)
+                else:
+                    gpu_pd = gpu_graph.build_partition(
+                        structure, P, r, cfg.cutoff, 0.0, 1e-8, False, dev)
+            else:
+                dist_info = self.build_graph(structure)
+        if gpu_pd is not None and P == 1:
+            pd = gpu_pd
+            plan = []
+            gids = np.arange(pd.n_atoms)
+            n_owned = pd.n_atoms
+        elif gpu_pd is not None:
+            pd = gpu_pd
+            plan = halo_plan(pd.markers, r, P)
+            gids = pd.global_ids
+            n_owned = pd.n_owned
+        else:
+            pd = PartitionData(dist_info, r, dev, use_bond_graph=False)
+            plan = halo_plan(dist_info.markers[r], r, P)
+            gids = np.asarray(dist_info.global_ids[r])
+            n_owned = dist_info.num_owned_atoms(r)
+        halo_seq = _HaloSeq()
+
+        def _halo(x):
+            if not plan:
+                return x
+            return HaloExchange.apply(_flat(x), plan,
+                                      halo_seq).view(-1, S, C)
+
+        lat0 = torch.tensor(np.asarray(structure.lattice), dtype=ft,
+                            device=dev)
+        strain = lat0.new_zeros(3, 3)
+        if calc_stresses:
+            strain.requires_grad_(True)
+        lattice = lat0 @ (torch.eye(3, device=dev, dtype=ft) + strain)
+        frac_local = torch.tensor(
+            np.asarray(structure.frac_coords)[gids], dtype=ft, device=dev)
+        pos = frac_local @ lattice
+        if not pos.requires_grad:
+            pos.requires_grad_(True)
+        pos.retain_grad()
+        spec = np.asarray(structure.species)
+        species = torch.tensor(spec[gids], dtype=torch.long, device=dev)
+
+        if gpu_pd is not None:
+            off_local = pd.off_i8.to(ft)
+        elif (csr := dist_info.csr_parts[r]
+              if getattr(dist_info, "csr_parts", None) else None) is not None:
+            off_local = torch.from_numpy(csr["offsets_i8"]).to(dev).to(ft)
+        else:
+            egids = np.asarray(dist_info.L2G_DE_mapping_list[r])
+            off_local = torch.tensor(np.asarray(dist_info.py_offsets)[egids],
+                                     dtype=ft, device=dev)
+
+        src_l, dst_l = pd.src.long(), pd.dst.long()
+        vectors = pos[dst_l] + off_local @ lattice - pos[src_l]
+        lengths = torch.linalg.norm(vectors, dim=1)
+
+        with torch.autocast("cuda", dtype=torch.bfloat16,
+                            enabled=self.autocast_bf16
+                            and dev.type == "cuda"):
+            R = so3.edge_align_rotation(vectors)
+            D = so3.wigner_D_batch(R, cfg.lmax)
+            Dinv = D.transpose(-1, -2)
+            x_edge = uma_ops.edge_scalars(core, lengths, species[src_l],
+                                          species[dst_l])
+
+            x = pos.new_zeros(len(species), S, C)
+            x = x.index_put(
+                (torch.arange(len(species), device=dev),
+                 torch.zeros(len(species), dtype=torch.long, device=dev)),
+                core.sphere_embedding[species])
+
+            # edge-degree embedding + scatter (escn_md.py:416 transfer)
+            ac = self.autocast_bf16 and dev.type == "cuda"
+
+            def _f32(t):
+                # the HIP scatter kernel is f32; under bf16 autocast the
+                # messages arrive bf16 — cast only then (never downcast
+                # the fp64 CPU test path)
+                return t.float() if ac else t
+
+            med = uma_ops.edge_degree_embed(core, x_edge, Dinv)
+            x = x + ops.scatter_edges(_f32(_flat(med)).contiguous(),
+                                      pd).view(-1, S, C) / cfg.avg_degree
+            x = _halo(x)
+
+            ckpt = self.checkpoint == "on" or (
+                self.checkpoint == "auto" and len(pd.src) > 4_000_000)
+            src_csr = (pd.src_perm, pd.src_row_ptr) \
+                if hasattr(pd, "src_perm") else None
+
+            for li, blk in enumerate(core.blocks):
+                def body(x, _blk=blk):
+                    h = uma_ops.rms_norm(x, _blk.norm1, cfg.lmax)
+                    hf = _flat(h).contiguous()
+                    x_src = ops.gather(hf, pd.src,
+                                       csr=src_csr).view(-1, S, C)
+                    x_dst = ops.gather(hf, pd.dst,
+                                       csr=(None, pd.row_ptr)
+                                       ).view(-1, S, C)
+                    msg = uma_ops.block_message(_blk, cfg, x_src, x_dst,
+                                                x_edge, D, Dinv)
+                    x2 = x + ops.scatter_edges(
+                        _f32(_flat(msg)).contiguous(), pd
+                    ).view(-1, S, C) / cfg.avg_degree
+                    return x2 + uma_ops.node_ffn(_blk, cfg, x2)
+
+                if ckpt:
+                    x = torch.utils.checkpoint.checkpoint(
+                        body, x, use_reentrant=False)
+                else:
+                    x = body(x)
+                if li < cfg.num_layers - 1:
+                    x = _halo(x)          # escn_md.py:496 transfer point
+
+            es = uma_ops.energy_head(core, _f32(x))
+
+        loss = es[:n_owned].sum()
+        grads = [pos, strain] if calc_stresses else [pos]
+        gv = torch.autograd.grad(loss, grads)
+        pos_grad = gv[0]
+        recvs = _exchange(pos_grad, plan, reverse=True)
+        pos_grad = pos_grad.clone()
+        for (q, ss, se, rs, re) in plan:
+            if se > ss:
+                pos_grad[ss:se] += recvs[q]
+        forces_owned = -pos_grad[:n_owned]
+
+        scal = loss.detach().clone()
+        if P > 1:
+            dist.all_reduce(scal)
+        out = {"energy": scal, "forces_owned": forces_owned,
+               "n_owned": n_owned, "global_ids_owned": gids[:n_owned]}
+        if calc_stresses:
+            sg = gv[1].detach().clone()
+            if P > 1:
+                dist.all_reduce(sg)
+            volume = float(np.abs(np.linalg.det(
+                np.asarray(structure.lattice))))
+            out["stress"] = -sg / volume * -160.21766208
+        return out
