@@ -93,7 +93,7 @@ def parse_args():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--workload", type=str, default="si1m",
                     choices=["li100k", "si1m", "si2m", "si1k",
-                             "mace62k", "mace500k"])
+                             "mace62k", "mace500k", "uma250k", "uma2m"])
     ap.add_argument("--threads", type=int, default=0,
                     help="graph-builder threads (0 = cpu_count/world)")
     ap.add_argument("--verlet", type=float, default=0.0, metavar="SKIN",
@@ -183,6 +183,27 @@ def cpu_baseline_leg(workload_name, threads):
     from distmlip_amd.structures import diamond_si
 
     torch.set_num_threads(threads)
+    if workload_name.startswith("uma"):
+        from distmlip_amd.uma_model import UMACore
+        from oracle.uma_ref import uma_oracle_forward
+
+        s = diamond_si(4, jitter=0.35, seed=0)   # 512-atom bounded sample
+        rng = _np.random.default_rng(99)
+        s.species = rng.integers(0, 3, size=s.num_atoms).astype(_np.int64)
+        t0 = time.time()
+        d = Distributed.create_distributed(
+            s.cart_coords, s.frac_coords, s.lattice, 1, s.pbc, 6.0, 0.0,
+            use_bond_graph=False, num_threads=threads)
+        core = UMACore.seeded(seed=0).float()
+        uma_oracle_forward(core, s, d.py_index_1, d.py_index_2,
+                           d.py_offsets, dtype=torch.float32)
+        dt = time.time() - t0
+        return {"value": s.num_atoms / dt, "unit": "atom_steps_per_s",
+                "cores": threads, "kind": "port",
+                "sample": f"diamond-Si {s.num_atoms} atoms, 1 full UMA "
+                          f"E+F forward (graph build + oracle "
+                          f"restatement), {dt:.2f}s"}
+
     if workload_name.startswith("mace"):
         from distmlip_amd.mace_model import MACECore
         from oracle.mace_ref import mace_oracle_forward
@@ -240,7 +261,8 @@ def main():
     from distmlip_amd.structures import workload
 
     is_mace = args.workload.startswith("mace")
-    weak = args.workload in ("li100k", "mace62k")
+    is_uma = args.workload.startswith("uma")
+    weak = args.workload in ("li100k", "mace62k", "uma250k")
     s = workload(args.workload, n_gpus=n_gpus if weak else 1)
 
     if world > 1:
@@ -248,7 +270,13 @@ def main():
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         torch.cuda.set_device(local_rank)
 
-    if is_mace:
+    if is_uma:
+        from distmlip_amd.uma_model import UMACore
+        from distmlip_amd.uma_runtime import UmaSpmdEngine
+        core = UMACore.seeded(seed=0).float()
+        engine = UmaSpmdEngine(core, world, threads=threads,
+                               autocast_bf16=True)
+    elif is_mace:
         from distmlip_amd.mace_model import MACECore
         from distmlip_amd.mace_runtime import MaceSpmdEngine
         core = MACECore.seeded(seed=0).float()
@@ -301,7 +329,7 @@ def main():
                       f"{len(v)} calls", file=sys.stderr)
         atexit.register(report)
 
-    timer = SegSumTimer(D=0 if is_mace else 64).wrap()
+    timer = SegSumTimer(D=0 if (is_mace or is_uma) else 64).wrap()
 
     graph_ms, model_ms = [], []
     for _ in range(args.warmup):
@@ -357,7 +385,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak" if weak else "strong",
             "vs_baseline": None,
-            "dtype": "f32",
+            "dtype": "bf16-autocast" if is_uma else "f32",
             "data": "synthetic",
             "config": {
                 "workload": args.workload,
@@ -365,8 +393,9 @@ def main():
                 "verlet_rebuilds": (getattr(engine, "_vcache", None) or
                                     {}).get("rebuilds"),
                 "n_atoms": int(total_atoms),
-                "model": "mace-mp-0-medium-shape" if is_mace
-                         else "chgnet-shape",
+                "model": ("uma-escn-shape" if is_uma else
+                          "mace-mp-0-medium-shape" if is_mace
+                          else "chgnet-shape"),
                 "cutoff": 6.0,
                 "three_body_cutoff": None if is_mace else 3.0,
                 "use_bond_graph": False if is_mace else use_bg,

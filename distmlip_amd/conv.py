@@ -517,7 +517,7 @@ def whole_graph_available(ops, mlp_pack) -> bool:
     """Whole-graph sequenced-reverse policy: single partition, frozen
     weights, raw-primitive backend, recording mode.  Default follows the
     measured A/B (DM_WHOLE_GRAPH=1/0 overrides)."""
-    default = "0"          # flipped only on a measured win (VERDICT #3)
+    default = "1"          # keep-mode measured win (runtime gates ckpt mode)
     return (os.environ.get("DM_WHOLE_GRAPH", default) == "1"
             and hasattr(ops, "r_gather_add3")
             and not mlp_pack[0].requires_grad

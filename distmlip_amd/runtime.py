@@ -503,7 +503,13 @@ class SpmdEngine:
         # verlet masks): every block + the final conv in ONE Function
         # whose backward is sequenced by hand (conv.py)
         from distmlip_amd.conv import _WholeGraphFn, whole_graph_available
+        # measured A/B (r2, same box): keep-mode wins at li100k (137.1 vs
+        # 140.5 ms/step); the streamed-recompute mode LOSES at si1m
+        # (2486 vs 2081 ms — the lean re-runs outweigh the saved adds), so
+        # checkpointed steps stay on the op-by-op path unless forced
+        wg_ckpt_ok = _os.environ.get("DM_WHOLE_GRAPH_CKPT") == "1"
         if (P == 1 and not mask_cutoffs and d % 2 == 0
+                and (not ckpt or wg_ckpt_ok)
                 and whole_graph_available(
                     ops, _packed_weights(core.atom_convs[0].edge_mlp))):
             wbb_m = w_bb if w_bb is not None else \

@@ -143,4 +143,19 @@ def workload(name: str, n_gpus: int = 1, seed: int = 0) -> Structure:
         rng = np.random.default_rng(seed + 77)
         s.species = rng.integers(0, 3, size=s.num_atoms).astype(np.int64)
         return s
+    if name == "uma2m":
+        # config #5 — UMA bf16, 2M-atom amorphous structure (8-GPU):
+        # jittered diamond-Si at amorphous-like disorder, 3 species
+        s = diamond_si(63, jitter=0.35, seed=seed)    # 2,000,376 atoms
+        rng = np.random.default_rng(seed + 99)
+        s.species = rng.integers(0, 3, size=s.num_atoms).astype(np.int64)
+        return s
+    if name == "uma250k":
+        # per-GPU unit of config #5 (2M/8), weak scaling with n_gpus
+        target = 2000376 // 8 * n_gpus
+        reps = int(round((target / 8) ** (1.0 / 3.0)))
+        s = diamond_si(reps, jitter=0.35, seed=seed)
+        rng = np.random.default_rng(seed + 99)
+        s.species = rng.integers(0, 3, size=s.num_atoms).astype(np.int64)
+        return s
     raise ValueError(f"unknown workload {name!r}")
