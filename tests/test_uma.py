@@ -125,6 +125,40 @@ def test_uma_engine_world1_vs_oracle():
     assert np.abs(out["stress"].numpy() - ref["stress"].numpy()).max() < 1e-7
 
 
+@pytest.mark.parametrize("P", [1, 2])
+def test_uma_dist_api_mirror_vs_oracle(P):
+    """The reference-shaped plugin surface (UMA_Dist.from_existing /
+    enable_distributed_mode / forward, escn_md.py:525-570 mirror)
+    against the oracle, P cpu partitions in one process."""
+    from distmlip_amd.dist import Distributed
+    from distmlip_amd.uma import UMA_Dist
+
+    s = diamond_si((8, 2, 2), jitter=0.15, seed=4)
+    s.species = np.asarray(s.species) % 3
+    core = _small_core(seed=11)
+
+    d = Distributed.create_distributed(
+        s.cart_coords, s.frac_coords, s.lattice, P, s.pbc, 6.0, 0.0,
+        use_bond_graph=False, num_threads=2)
+    model = UMA_Dist.from_existing(core)
+    model.enable_distributed_mode(["cpu"] * P)
+
+    pos = torch.tensor(s.frac_coords @ s.lattice, dtype=torch.float64,
+                       requires_grad=True)
+    shifts = torch.tensor(np.asarray(d.py_offsets) @ s.lattice,
+                          dtype=torch.float64)
+    data = {"positions": pos,
+            "species": torch.tensor(np.asarray(s.species),
+                                    dtype=torch.long),
+            "shifts": shifts}
+    out = model.forward(data, d)
+    ref = uma_oracle_forward(core, s, d.py_index_1, d.py_index_2,
+                             d.py_offsets)
+    assert abs(out["energy"].item() - ref["energy"].item()) < 1e-9
+    dF = (out["forces"] - ref["forces"]).abs().max().item()
+    assert dF < 5e-9, dF
+
+
 def _worker(rank, world, init_file, out_dir):
     import sys
     sys.path.insert(0, os.path.dirname(os.path.dirname(
