@@ -21,14 +21,17 @@ def _grids(cfg: UMAConfig, like: torch.Tensor):
 
 
 def rms_norm(x: torch.Tensor, scale: torch.Tensor, lmax: int,
-             eps: float = 1e-8) -> torch.Tensor:
-    """Equivariant RMS norm: per (l, channel), normalize by the RMS over
-    the 2l+1 m-components; learned scale per (l, channel)."""
+             eps: float = 1e-6) -> torch.Tensor:
+    """Equivariant RMS norm per (node, l) over the (m, channel) block
+    jointly (the eSCN norm family's shape); learned scale per
+    (l, channel).  Normalizing per channel over only the 2l+1
+    m-components makes near-zero channels explode through rsqrt —
+    measured as ~1e5-magnitude forces at the full shape."""
     outs = []
     for l in range(lmax + 1):
         o, d = l * l, 2 * l + 1
         blk = x[:, o:o + d, :]
-        ms = blk.pow(2).mean(dim=1, keepdim=True)
+        ms = blk.pow(2).mean(dim=(1, 2), keepdim=True)
         outs.append(blk * torch.rsqrt(ms + eps) * scale[l])
     return torch.cat(outs, dim=1)
 
@@ -57,13 +60,22 @@ def so2_conv(conv: SO2Conv, xt: torch.Tensor, gate: torch.Tensor,
     return out
 
 
-def s2_act(x: torch.Tensor, cfg: UMAConfig) -> torch.Tensor:
+def s2_act(x: torch.Tensor, cfg: UMAConfig,
+           chunk: int = 65536) -> torch.Tensor:
     """Pointwise silu on the sphere: to grid, silu, band-limited
-    projection back (the eSCN grid activation)."""
+    projection back (the eSCN grid activation).  Chunked over nodes —
+    the [N, G, C] grid tensor at 238k atoms x 512 points x 128 channels
+    is 62 GB fp32, so it is never materialized whole."""
     to_g, from_g = _grids(cfg, x)
-    f = torch.einsum("gs,nsc->ngc", to_g, x)
-    f = torch.nn.functional.silu(f)
-    return torch.einsum("sg,ngc->nsc", from_g, f)
+
+    def one(t):
+        f = torch.nn.functional.silu(torch.einsum("gs,nsc->ngc", to_g, t))
+        return torch.einsum("sg,ngc->nsc", from_g, f)
+
+    if x.shape[0] <= chunk:
+        return one(x)
+    return torch.cat([one(x[i:i + chunk])
+                      for i in range(0, x.shape[0], chunk)], dim=0)
 
 
 def edge_scalars(core: UMACore, lengths: torch.Tensor, spec_src,

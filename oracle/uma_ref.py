@@ -49,12 +49,12 @@ def _pinv_grids(cfg, dtype):
     return Y, from_g
 
 
-def _norm(x, scale, lmax, eps=1e-8):
+def _norm(x, scale, lmax, eps=1e-6):
     outs = []
     for l in range(lmax + 1):
         o, d = l * l, 2 * l + 1
         blk = x[:, o:o + d, :]
-        rms = (blk.pow(2).mean(1, keepdim=True) + eps).sqrt()
+        rms = (blk.pow(2).mean((1, 2), keepdim=True) + eps).sqrt()
         outs.append(blk / rms * scale[l])
     return torch.cat(outs, 1)
 

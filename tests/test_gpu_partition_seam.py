@@ -92,9 +92,14 @@ def test_gpu_build_partition_matches_native(P):
     markers bit-equal (set-determined); edge/bond/line layers equal as
     sets (the GPU NL's per-dst row order may differ from the CPU
     builder's global emission order — physics-identical, and
-    rank-consistent since every rank builds from the same full NL)."""
-    reps = (16, 2, 2) if P >= 4 else (12, 2, 2)
+    rank-consistent since every rank builds from the same full NL).
+
+    Cells must satisfy gpu_graph.supported (>= 3 cells of >= cutoff per
+    dim) — the engines guard this; a first run of this test with 11 A
+    transverse dims silently under-binned the cell list."""
+    reps = (16, 4, 4) if P >= 4 else (12, 4, 4)
     s = diamond_si(reps, jitter=0.12, seed=2)
+    assert gpu_graph.supported(s, 6.0)
     dev = torch.device("cuda:0")
     for r in range(P):
         pd = gpu_graph.build_partition(s, P, r, 6.0, 3.0, 1e-8, True, dev)
