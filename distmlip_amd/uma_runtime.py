@@ -11,6 +11,7 @@ reductions in fp32) — the BASELINE config #5 dtype.
 """
 from __future__ import annotations
 
+import os as _os
 from copy import deepcopy
 from typing import Optional
 
@@ -159,9 +160,21 @@ class UmaSpmdEngine:
                 # the fp64 CPU test path)
                 return t.float() if ac else t
 
-            med = uma_ops.edge_degree_embed(core, x_edge, Dinv)
-            x = x + ops.scatter_edges(_f32(_flat(med)).contiguous(),
-                                      pd).view(-1, S, C) / cfg.avg_degree
+            E_all = len(pd.src)
+            if E_all > 2_000_000:
+                accd = torch.zeros_like(x)
+                step_e = 2_000_000
+                for e0 in range(0, E_all, step_e):
+                    e1 = min(e0 + step_e, E_all)
+                    md = uma_ops.edge_degree_embed(core, x_edge[e0:e1],
+                                                   Dinv[e0:e1])
+                    accd = accd.index_add(0, dst_l[e0:e1], md)
+                x = x + accd / cfg.avg_degree
+            else:
+                med = uma_ops.edge_degree_embed(core, x_edge, Dinv)
+                x = x + ops.scatter_edges(
+                    _f32(_flat(med)).contiguous(), pd
+                ).view(-1, S, C) / cfg.avg_degree
             x = _halo(x)
 
             ckpt = self.checkpoint == "on" or (
@@ -169,20 +182,46 @@ class UmaSpmdEngine:
             src_csr = (pd.src_perm, pd.src_row_ptr) \
                 if hasattr(pd, "src_perm") else None
 
+            # per-edge rotated tensors ([E, S, 2C]) reach ~100 GB at 11M
+            # edges — the message pass runs over contiguous node-range
+            # chunks (dst-sorted edges), bounding transients
+            chunk_edges = int(_os.environ.get("DM_UMA_CHUNK", 1_500_000))
+            E_tot = len(pd.src)
+            if E_tot > chunk_edges:
+                rp = pd.row_ptr.long().cpu().numpy()
+                ranges = []
+                n0 = 0
+                for n in range(1, len(rp)):
+                    if rp[n] - rp[n0] >= chunk_edges or n == len(rp) - 1:
+                        ranges.append((int(rp[n0]), int(rp[n])))
+                        n0 = n
+            else:
+                ranges = [(0, E_tot)]
+
             for li, blk in enumerate(core.blocks):
                 def body(x, _blk=blk):
                     h = uma_ops.rms_norm(x, _blk.norm1, cfg.lmax)
-                    hf = _flat(h).contiguous()
-                    x_src = ops.gather(hf, pd.src,
-                                       csr=src_csr).view(-1, S, C)
-                    x_dst = ops.gather(hf, pd.dst,
-                                       csr=(None, pd.row_ptr)
-                                       ).view(-1, S, C)
-                    msg = uma_ops.block_message(_blk, cfg, x_src, x_dst,
-                                                x_edge, D, Dinv)
-                    x2 = x + ops.scatter_edges(
-                        _f32(_flat(msg)).contiguous(), pd
-                    ).view(-1, S, C) / cfg.avg_degree
+                    if len(ranges) == 1:
+                        hf = _flat(h).contiguous()
+                        x_src = ops.gather(hf, pd.src,
+                                           csr=src_csr).view(-1, S, C)
+                        x_dst = ops.gather(hf, pd.dst,
+                                           csr=(None, pd.row_ptr)
+                                           ).view(-1, S, C)
+                        msg = uma_ops.block_message(_blk, cfg, x_src,
+                                                    x_dst, x_edge, D, Dinv)
+                        x2 = x + ops.scatter_edges(
+                            _f32(_flat(msg)).contiguous(), pd
+                        ).view(-1, S, C) / cfg.avg_degree
+                    else:
+                        acc = torch.zeros_like(x)
+                        for (e0, e1) in ranges:
+                            m = uma_ops.block_message(
+                                _blk, cfg, h[src_l[e0:e1]],
+                                h[dst_l[e0:e1]], x_edge[e0:e1],
+                                D[e0:e1], Dinv[e0:e1])
+                            acc = acc.index_add(0, dst_l[e0:e1], m)
+                        x2 = x + acc / cfg.avg_degree
                     return x2 + uma_ops.node_ffn(_blk, cfg, x2)
 
                 if ckpt:
