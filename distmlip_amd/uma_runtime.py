@@ -168,7 +168,7 @@ class UmaSpmdEngine:
                     e1 = min(e0 + step_e, E_all)
                     md = uma_ops.edge_degree_embed(core, x_edge[e0:e1],
                                                    Dinv[e0:e1])
-                    accd = accd.index_add(0, dst_l[e0:e1], md)
+                    accd = accd.index_add(0, dst_l[e0:e1], _f32(md))
                 x = x + accd / cfg.avg_degree
             else:
                 med = uma_ops.edge_degree_embed(core, x_edge, Dinv)
@@ -220,7 +220,7 @@ class UmaSpmdEngine:
                                 _blk, cfg, h[src_l[e0:e1]],
                                 h[dst_l[e0:e1]], x_edge[e0:e1],
                                 D[e0:e1], Dinv[e0:e1])
-                            acc = acc.index_add(0, dst_l[e0:e1], m)
+                            acc = acc.index_add(0, dst_l[e0:e1], _f32(m))
                         x2 = x + acc / cfg.avg_degree
                     return x2 + uma_ops.node_ffn(_blk, cfg, x2)
 
