@@ -74,8 +74,17 @@ def s2_act(x: torch.Tensor, cfg: UMAConfig,
 
     if x.shape[0] <= chunk:
         return one(x)
-    return torch.cat([one(x[i:i + chunk])
-                      for i in range(0, x.shape[0], chunk)], dim=0)
+    outs = []
+    for i in range(0, x.shape[0], chunk):
+        xc = x[i:i + chunk]
+        if torch.is_grad_enabled() and x.requires_grad:
+            # per-chunk checkpoint: silu saves its [N_c, G, C] grid input
+            # otherwise — ~8 GB per chunk held across the whole backward
+            outs.append(torch.utils.checkpoint.checkpoint(
+                one, xc, use_reentrant=False))
+        else:
+            outs.append(one(xc))
+    return torch.cat(outs, dim=0)
 
 
 def edge_scalars(core: UMACore, lengths: torch.Tensor, spec_src,

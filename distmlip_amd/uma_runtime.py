@@ -216,10 +216,22 @@ class UmaSpmdEngine:
                     else:
                         acc = torch.zeros_like(x)
                         for (e0, e1) in ranges:
-                            m = uma_ops.block_message(
-                                _blk, cfg, h[src_l[e0:e1]],
-                                h[dst_l[e0:e1]], x_edge[e0:e1],
-                                D[e0:e1], Dinv[e0:e1])
+                            # nested checkpoint: without it the recompute
+                            # backward holds EVERY chunk's [E_c, S, 2C]
+                            # rotation graph at once (~20 GB x n_chunks)
+                            def _chunk(hs, hd, xe, Dc, Dic, _b=_blk):
+                                return uma_ops.block_message(
+                                    _b, cfg, hs, hd, xe, Dc, Dic)
+                            if torch.is_grad_enabled():
+                                m = torch.utils.checkpoint.checkpoint(
+                                    _chunk, h[src_l[e0:e1]],
+                                    h[dst_l[e0:e1]], x_edge[e0:e1],
+                                    D[e0:e1], Dinv[e0:e1],
+                                    use_reentrant=False)
+                            else:
+                                m = _chunk(h[src_l[e0:e1]],
+                                           h[dst_l[e0:e1]], x_edge[e0:e1],
+                                           D[e0:e1], Dinv[e0:e1])
                             acc = acc.index_add(0, dst_l[e0:e1], _f32(m))
                         x2 = x + acc / cfg.avg_degree
                     return x2 + uma_ops.node_ffn(_blk, cfg, x2)
