@@ -111,23 +111,80 @@ def conv_tp_messages(inter: Interaction, gathered: Dict[int, torch.Tensor],
     the k1-contraction (d1 <= 3) is unrolled into broadcast
     multiply-accumulates, which are plain HBM-bound elementwise
     kernels."""
-    msgs: Dict[int, torch.Tensor] = {}
+    # One tall GEMM per sender-l: with z[e,c,(k1,k2)] = x[e,c,k1] Y[e,k2]
+    # (the x (x) Y outer product), every path with this l1 is
+    #   u_p[e,c,k3] = sum_{k1,k2} z[e,c,(k1,k2)] C_p[k1,k2,k3]
+    # so stacking the C_p blocks column-wise gives u_all = z @ Cbig —
+    # [E*C, d1*16] x [d1*16, sum_p d3], an MFMA-shaped GEMM.  Per-path
+    # weighting and per-l3 accumulation are then a handful of broadcast
+    # ops.  (The per-path unrolled form spawned ~60 broadcast muls whose
+    # backwards are reduction kernels — measured 57% of the step,
+    # profiles/r2_mace_kernel_stats_postfix.csv.)
+    import os
+    E = Y.shape[0]
+    C = next(iter(gathered.values())).shape[1]
+    by_l1: Dict[int, list] = {}
     for p, (l1, l2, l3) in enumerate(inter.paths):
-        CG = cg_t(l1, l2, l3, Y)
+        by_l1.setdefault(l1, []).append((p, l2, l3))
+    l3s = sorted({l3 for (_, _, l3) in inter.paths})
+
+    def tp_chunk(Y_c, w_c, *g_c):
+        out = {}
+        Ec = Y_c.shape[0]
+        for gi, (l1, plist) in enumerate(by_l1.items()):
+            d1 = so3.L_DIMS[l1]
+            Cbig = _tp_cbig(l1, tuple(plist),
+                            str(Y_c.dtype).split(".")[-1], str(Y_c.device))
+            z = (g_c[gi].unsqueeze(-1) * Y_c.view(Ec, 1, 1, 16)
+                 ).reshape(Ec, C, d1 * 16)
+            u = z @ Cbig                              # [Ec, C, sum d3]
+            col = 0
+            for (p, l2, l3) in plist:
+                d3 = so3.L_DIMS[l3]
+                m = u[:, :, col:col + d3] * w_c[:, p, :].unsqueeze(-1)
+                col += d3
+                out[l3] = out[l3] + m if l3 in out else m
+        return tuple(out[l3] for l3 in l3s)
+
+    chunk = int(os.environ.get("DM_MACE_TP_CHUNK", 1_000_000))
+    if E <= chunk:
+        parts = [tp_chunk(Y, tp_w, *(gathered[l1] for l1 in by_l1))]
+    else:
+        # chunk edges; checkpoint each chunk so the z/u transients
+        # (z is [E, C, 48] = 71 GB unchunked at 2.9M edges) are
+        # recomputed per chunk during backward, never all live at once
+        parts = []
+        for e0 in range(0, E, chunk):
+            e1 = min(e0 + chunk, E)
+            args = (Y[e0:e1], tp_w[e0:e1],
+                    *(gathered[l1][e0:e1] for l1 in by_l1))
+            if torch.is_grad_enabled() and Y.requires_grad:
+                parts.append(torch.utils.checkpoint.checkpoint(
+                    tp_chunk, *args, use_reentrant=False))
+            else:
+                parts.append(tp_chunk(*args))
+    return {l3: torch.cat([p[i] for p in parts], dim=0)
+            for i, l3 in enumerate(l3s)}
+
+
+@lru_cache(maxsize=None)
+def _tp_cbig(l1: int, plist, dtype_str: str, device_str: str):
+    """[d1*16, sum_p d3]: per path (l1,l2,l3), the CG block embedded at
+    k2-rows of the l2 block, stacked column-wise in path order."""
+    d1 = so3.L_DIMS[l1]
+    cols = sum(so3.L_DIMS[l3] for (_, _, l3) in plist)
+    Cbig = np.zeros((d1 * 16, cols))
+    col = 0
+    for (_, l2, l3) in plist:
+        CG = so3.real_cg(l1, l2, l3)                  # [d1, d2, d3]
         o2, d2 = so3.L_OFF[l2], so3.L_DIMS[l2]
-        d1, d3 = so3.L_DIMS[l1], so3.L_DIMS[l3]
-        yb = Y[:, o2:o2 + d2]
-        # [E, d2] @ [d2, d1*d3] -> [E, d1, d3]
-        yc = (yb @ CG.permute(1, 0, 2).reshape(d2, d1 * d3)).view(-1, d1, d3)
-        xa = gathered[l1] * tp_w[:, p, :].unsqueeze(-1)    # [E, C, d1]
-        m = xa[:, :, 0:1] * yc[:, 0, :].unsqueeze(1)
-        for k1 in range(1, d1):
-            m = m + xa[:, :, k1:k1 + 1] * yc[:, k1, :].unsqueeze(1)
-        if l3 in msgs:
-            msgs[l3] = msgs[l3] + m
-        else:
-            msgs[l3] = m
-    return msgs
+        d3 = so3.L_DIMS[l3]
+        blk = np.zeros((d1, 16, d3))
+        blk[:, o2:o2 + d2, :] = CG
+        Cbig[:, col:col + d3] = blk.reshape(d1 * 16, d3)
+        col += d3
+    return torch.tensor(Cbig, dtype=getattr(torch, dtype_str),
+                        device=torch.device(device_str))
 
 
 @lru_cache(maxsize=None)
