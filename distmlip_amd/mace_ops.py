@@ -104,22 +104,41 @@ def conv_tp_messages(inter: Interaction, gathered: Dict[int, torch.Tensor],
     path weights (radial MLP output).  Returns per-l3 messages
     [E, C, d3].
 
-    Shape discipline (measured, profiles/r2_mace_kernel_stats.csv): a
-    batch-E bmm of (C x d1)(d1 x d3) tiles runs rocBLAS at ~1% of peak
-    (MT16x16 strided-batched tiles, ~100 ms per call at E=2.9M) — 96% of
-    a 46 s step.  yc stays a tall-skinny GEMM ([E,d2] x [d2, d1*d3]);
-    the k1-contraction (d1 <= 3) is unrolled into broadcast
-    multiply-accumulates, which are plain HBM-bound elementwise
-    kernels."""
-    # One tall GEMM per sender-l: with z[e,c,(k1,k2)] = x[e,c,k1] Y[e,k2]
-    # (the x (x) Y outer product), every path with this l1 is
-    #   u_p[e,c,k3] = sum_{k1,k2} z[e,c,(k1,k2)] C_p[k1,k2,k3]
-    # so stacking the C_p blocks column-wise gives u_all = z @ Cbig —
-    # [E*C, d1*16] x [d1*16, sum_p d3], an MFMA-shaped GEMM.  Per-path
-    # weighting and per-l3 accumulation are then a handful of broadcast
-    # ops.  (The per-path unrolled form spawned ~60 broadcast muls whose
-    # backwards are reduction kernels — measured 57% of the step,
-    # profiles/r2_mace_kernel_stats_postfix.csv.)
+    Formulation policy (measured, mace62k same box):
+      * batch-E bmm of (C x d1)(d1 x d3) tiles: rocBLAS MT16x16 at ~1%
+        of peak, 46.4 s/step — never used;
+      * per-path broadcast multiply-accumulates (default): 1.67 s/step;
+      * per-l1 z-outer + stacked-CG tall GEMM with chunked
+        checkpointing (DM_MACE_TP=zouter): 2.46 s/step — the recompute
+        and the [E,C,d1*16] z traffic outweigh the saved reduction
+        kernels at these shapes; kept for experimentation and as the
+        stepping stone to the fused HIP TP kernel.
+    """
+    import os
+    if os.environ.get("DM_MACE_TP") == "zouter":
+        return _conv_tp_zouter(inter, gathered, Y, tp_w)
+    msgs: Dict[int, torch.Tensor] = {}
+    for p, (l1, l2, l3) in enumerate(inter.paths):
+        CG = cg_t(l1, l2, l3, Y)
+        o2, d2 = so3.L_OFF[l2], so3.L_DIMS[l2]
+        d1, d3 = so3.L_DIMS[l1], so3.L_DIMS[l3]
+        yb = Y[:, o2:o2 + d2]
+        # [E, d2] @ [d2, d1*d3] -> [E, d1, d3]
+        yc = (yb @ CG.permute(1, 0, 2).reshape(d2, d1 * d3)).view(-1, d1, d3)
+        xa = gathered[l1] * tp_w[:, p, :].unsqueeze(-1)    # [E, C, d1]
+        m = xa[:, :, 0:1] * yc[:, 0, :].unsqueeze(1)
+        for k1 in range(1, d1):
+            m = m + xa[:, :, k1:k1 + 1] * yc[:, k1, :].unsqueeze(1)
+        if l3 in msgs:
+            msgs[l3] = msgs[l3] + m
+        else:
+            msgs[l3] = m
+    return msgs
+
+
+def _conv_tp_zouter(inter: Interaction, gathered: Dict[int, torch.Tensor],
+                    Y: torch.Tensor, tp_w: torch.Tensor
+                    ) -> Dict[int, torch.Tensor]:
     import os
     E = Y.shape[0]
     C = next(iter(gathered.values())).shape[1]
