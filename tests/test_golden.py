@@ -70,3 +70,40 @@ def test_golden_reference_partition_shapes(ref_graph_backend):
     assert [np.asarray(m).tolist() for m in ref[12]] == exp["line_markers"]
     assert [int(len(np.asarray(e))) for e in ref[0]] == exp["edges_per_partition"]
     assert [int(len(np.asarray(e))) for e in ref[9]] == exp["lines_per_partition"]
+
+
+def test_golden_mace_uma():
+    """Round-2 model families pinned by committed fixtures
+    (tests/golden/make_golden_models.py)."""
+    from distmlip_amd.mace_model import MACEConfig, MACECore
+    from distmlip_amd.uma_model import UMAConfig, UMACore
+    from oracle.mace_ref import mace_oracle_forward
+    from oracle.uma_ref import uma_oracle_forward
+
+    with open(os.path.join(os.path.dirname(__file__), "golden",
+                           "golden_models.json")) as f:
+        g0 = json.load(f)
+    s = diamond_si((6, 2, 2), jitter=0.1, seed=2)
+    s.species = np.asarray(s.species) % 3
+    g = brute_force_neighbors(s.frac_coords, s.lattice, s.pbc, 6.0, 0.0)
+
+    mc = MACECore.seeded(MACEConfig(n_elements=3, channels=16,
+                                    avg_num_neighbors=20.0,
+                                    atomic_inter_scale=0.7,
+                                    atomic_inter_shift=0.1),
+                         seed=9).double()
+    rm = mace_oracle_forward(mc, s, g["src"], g["dst"], g["offsets"])
+    assert abs(rm["energy"].item() - g0["mace"]["energy"]) < 1e-9
+    np.testing.assert_allclose(rm["forces"][:5].numpy(),
+                               np.asarray(g0["mace"]["forces_head"]),
+                               rtol=1e-9, atol=1e-10)
+
+    uc = UMACore.seeded(UMAConfig(n_elements=3, sphere_channels=16,
+                                  num_layers=2, edge_ch=32, num_gauss=16,
+                                  spec_emb=8, avg_degree=20.0),
+                        seed=9).double()
+    ru = uma_oracle_forward(uc, s, g["src"], g["dst"], g["offsets"])
+    assert abs(ru["energy"].item() - g0["uma"]["energy"]) < 1e-9
+    np.testing.assert_allclose(ru["forces"][:5].numpy(),
+                               np.asarray(g0["uma"]["forces_head"]),
+                               rtol=1e-9, atol=1e-10)
