@@ -17,6 +17,10 @@ Workloads (BASELINE.json configs; distmlip_amd/structures.py):
   li100k          : BCC-Li ~100k atoms PER GPU (weak scaling; N=1 is
       exactly config #2, 101,306 atoms)
   si1k            : 1,000-atom plumbing config (#1)
+  mace62k/mace500k: config #4 (MACE-MP-0-medium shape) — 500k/8 atoms
+      per GPU weak unit / 512k fixed
+  uma250k/uma2m   : config #5 (UMA eSCN shape, bf16 autocast) — 2M/8
+      atoms per GPU weak unit / 2.0M fixed
 """
 from __future__ import annotations
 
