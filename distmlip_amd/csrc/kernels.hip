@@ -614,6 +614,202 @@ __global__ void k_nl(const double* __restrict__ pos,
 
 }  // namespace
 
+
+// ---------------------------------------------------------------------------
+// MACE uvu tensor product, fused per edge (round 2; the equivariant
+// per-edge message contraction of the MACE interactions — reference
+// delegates it to e3nn TensorProduct, implementations/mace/models.py:144-152).
+// One 64-lane wave per edge, channels split across lanes (CPL = C/64 per
+// lane).  CG nonzeros stream through the SCALAR cache (wave-uniform
+// entries, sorted by path so the per-path weight row loads once);
+// k-indexed accumulators live in LDS (dynamic register indexing would
+// spill to scratch).  Everything HBM-side is lane-contiguous.
+// ---------------------------------------------------------------------------
+
+template <int CPL>
+__global__ void k_mace_tp_fwd(const float* __restrict__ x0,
+                              const float* __restrict__ x1,
+                              const float* __restrict__ Y,
+                              const float* __restrict__ w,
+                              const int32_t* __restrict__ nz,
+                              const float* __restrict__ nzc,
+                              int32_t nnz,
+                              float* __restrict__ o0,
+                              float* __restrict__ o1,
+                              float* __restrict__ o2,
+                              float* __restrict__ o3,
+                              int64_t E, int32_t C, int32_t P,
+                              int32_t d1b) {
+    __shared__ float acc[CPL * 16 * 4 * 64];     // [i][k3][wave][lane]
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+    for (int64_t e = blockIdx.x * (int64_t)nw + wid; e < E;
+         e += (int64_t)gridDim.x * nw) {
+        float xr0[CPL];
+        float xr1[CPL > 0 ? CPL : 1][3];
+        for (int i = 0; i < CPL; ++i)
+            xr0[i] = x0[e * C + lane + i * 64];
+        if (x1 != nullptr)
+            for (int i = 0; i < CPL; ++i)
+                for (int k = 0; k < 3; ++k)
+                    xr1[i][k] = (k < d1b)
+                        ? x1[(e * C + lane + i * 64) * d1b + k] : 0.0f;
+        for (int i = 0; i < CPL; ++i)
+            for (int k3 = 0; k3 < 16; ++k3)
+                acc[((i * 16 + k3) * 4 + wid) * 64 + lane] = 0.0f;
+        const float* Ye = Y + e * 16;
+        int curp = -1;
+        float wreg[CPL];
+        for (int32_t t = 0; t < nnz; ++t) {
+            const int32_t p = nz[5 * t], slot = nz[5 * t + 1],
+                          k1 = nz[5 * t + 2], k2 = nz[5 * t + 3],
+                          k3 = nz[5 * t + 4];
+            if (p != curp) {
+                curp = p;
+                for (int i = 0; i < CPL; ++i)
+                    wreg[i] = w[((int64_t)e * P + p) * C + lane + i * 64];
+            }
+            const float cy = nzc[t] * Ye[k2];
+            for (int i = 0; i < CPL; ++i) {
+                float xv;                      // k1 is wave-uniform:
+                if (slot == 0) xv = xr0[i];    // scalar-predicated chain,
+                else if (k1 == 0) xv = xr1[i][0];   // no scratch spill
+                else if (k1 == 1) xv = xr1[i][1];
+                else xv = xr1[i][2];
+                acc[((i * 16 + k3) * 4 + wid) * 64 + lane]
+                    += cy * wreg[i] * xv;
+            }
+        }
+        // k3 -> (l3 block, row) writeback into the four per-l3 tensors
+        // ([E, d3, C] each — contiguous rows for the scatter kernel)
+        float* const outs[4] = {o0, o1, o2, o3};
+        static const int BLK[16] = {0,1,1,1,2,2,2,2,2,3,3,3,3,3,3,3};
+        static const int ROW[16] = {0,0,1,2,0,1,2,3,4,0,1,2,3,4,5,6};
+        static const int DD[4] = {1,3,5,7};
+#pragma unroll
+        for (int k3 = 0; k3 < 16; ++k3)
+            for (int i = 0; i < CPL; ++i)
+                outs[BLK[k3]][((int64_t)e * DD[BLK[k3]] + ROW[k3]) * C
+                              + lane + i * 64] =
+                    acc[((i * 16 + k3) * 4 + wid) * 64 + lane];
+    }
+}
+
+template <int CPL>
+__global__ void k_mace_tp_bwd(const float* __restrict__ g0,
+                              const float* __restrict__ g1,
+                              const float* __restrict__ g2,
+                              const float* __restrict__ g3,
+                              const float* __restrict__ x0,
+                              const float* __restrict__ x1,
+                              const float* __restrict__ Y,
+                              const float* __restrict__ w,
+                              const int32_t* __restrict__ nz,
+                              const float* __restrict__ nzc,
+                              int32_t nnz,
+                              float* __restrict__ dx0,
+                              float* __restrict__ dx1,
+                              float* __restrict__ dY,
+                              float* __restrict__ dw,
+                              int64_t E, int32_t C, int32_t P,
+                              int32_t d1b) {
+    // LDS: go rows (dynamic k3 reads) + per-lane dY partials (dynamic k2)
+    __shared__ float gob[CPL * 16 * 4 * 64];
+    __shared__ float dyb[16 * 4 * 64];
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+    for (int64_t e = blockIdx.x * (int64_t)nw + wid; e < E;
+         e += (int64_t)gridDim.x * nw) {
+        float xr0[CPL];
+        float xr1[CPL > 0 ? CPL : 1][3];
+        float dxr0[CPL] = {};
+        float dxr1[CPL > 0 ? CPL : 1][3] = {};
+        for (int i = 0; i < CPL; ++i)
+            xr0[i] = x0[e * C + lane + i * 64];
+        if (x1 != nullptr)
+            for (int i = 0; i < CPL; ++i)
+                for (int k = 0; k < 3; ++k)
+                    xr1[i][k] = (k < d1b)
+                        ? x1[(e * C + lane + i * 64) * d1b + k] : 0.0f;
+        {
+            const float* const gos[4] = {g0, g1, g2, g3};
+            static const int BLK[16] = {0,1,1,1,2,2,2,2,2,3,3,3,3,3,3,3};
+            static const int ROW[16] = {0,0,1,2,0,1,2,3,4,0,1,2,3,4,5,6};
+            static const int DD[4] = {1,3,5,7};
+#pragma unroll
+            for (int k3 = 0; k3 < 16; ++k3)
+                for (int i = 0; i < CPL; ++i)
+                    gob[((i * 16 + k3) * 4 + wid) * 64 + lane] =
+                        gos[BLK[k3]][((int64_t)e * DD[BLK[k3]] + ROW[k3])
+                                     * C + lane + i * 64];
+        }
+        for (int k2 = 0; k2 < 16; ++k2)
+            dyb[(k2 * 4 + wid) * 64 + lane] = 0.0f;
+        const float* Ye = Y + e * 16;
+        int curp = -1;
+        float wreg[CPL];
+        float dwacc[CPL] = {};
+        for (int32_t t = 0; t < nnz; ++t) {
+            const int32_t p = nz[5 * t], slot = nz[5 * t + 1],
+                          k1 = nz[5 * t + 2], k2 = nz[5 * t + 3],
+                          k3 = nz[5 * t + 4];
+            if (p != curp) {
+                if (curp >= 0)
+                    for (int i = 0; i < CPL; ++i) {
+                        dw[((int64_t)e * P + curp) * C + lane + i * 64] =
+                            dwacc[i];
+                        dwacc[i] = 0.0f;
+                    }
+                curp = p;
+                for (int i = 0; i < CPL; ++i)
+                    wreg[i] = w[((int64_t)e * P + p) * C + lane + i * 64];
+            }
+            const float c = nzc[t];
+            const float yv = Ye[k2];
+            for (int i = 0; i < CPL; ++i) {
+                const float g = gob[((i * 16 + k3) * 4 + wid) * 64 + lane];
+                float xv;
+                if (slot == 0) xv = xr0[i];
+                else if (k1 == 0) xv = xr1[i][0];
+                else if (k1 == 1) xv = xr1[i][1];
+                else xv = xr1[i][2];
+                const float cg = c * g;
+                // d x
+                const float dxv = cg * yv * wreg[i];
+                if (slot == 0) dxr0[i] += dxv;
+                else if (k1 == 0) dxr1[i][0] += dxv;
+                else if (k1 == 1) dxr1[i][1] += dxv;
+                else dxr1[i][2] += dxv;
+                // d w
+                dwacc[i] += cg * yv * xv;
+                // d Y (per-lane partial; reduced after the loop)
+                dyb[(k2 * 4 + wid) * 64 + lane] += cg * wreg[i] * xv;
+            }
+        }
+        if (curp >= 0)
+            for (int i = 0; i < CPL; ++i)
+                dw[((int64_t)e * P + curp) * C + lane + i * 64] = dwacc[i];
+        for (int i = 0; i < CPL; ++i)
+            dx0[e * C + lane + i * 64] = dxr0[i];
+        if (dx1 != nullptr)
+            for (int i = 0; i < CPL; ++i)
+                for (int k = 0; k < d1b; ++k)
+                    dx1[(e * C + lane + i * 64) * d1b + k] = dxr1[i][k];
+        // reduce dY partials across the wave: waves are lockstep, each
+        // owns its LDS stripe — no block barrier (one inside this
+        // grid-stride loop would deadlock when wave trip counts differ)
+        if (lane < 16) {
+            float s = 0.0f;
+            for (int l2 = 0; l2 < 64; ++l2)
+                s += dyb[(lane * 4 + wid) * 64 + l2];
+            dY[e * 16 + lane] = s;
+        }
+    }
+}
+
+
 // ---------------------------------------------------------------------------
 // C ABI
 // ---------------------------------------------------------------------------
@@ -849,6 +1045,44 @@ int dm_nl_fill_f64(const double* pos, const int32_t* cid,
     k_nl<true><<<nblocks(N, BLOCK), BLOCK, 0, s>>>(
         pos, cid, order, cell_start, ncx, ncy, ncz, lx, ly, lz, r2tol, tol,
         br2tol, row_ptr, src, off_i8, bond_flag, N);
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+
+int dm_mace_tp_fwd_f32(const float* x0, const float* x1, const float* Y,
+                       const float* w, const int32_t* nz, const float* nzc,
+                       int32_t nnz, float* o0, float* o1, float* o2,
+                       float* o3, int64_t E, int32_t C,
+                       int32_t P, int32_t d1b, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (C == 64)
+        k_mace_tp_fwd<1><<<nblocks(E, 4), BLOCK, 0, s>>>(
+            x0, x1, Y, w, nz, nzc, nnz, o0, o1, o2, o3, E, C, P, d1b);
+    else if (C == 128)
+        k_mace_tp_fwd<2><<<nblocks(E, 4), BLOCK, 0, s>>>(
+            x0, x1, Y, w, nz, nzc, nnz, o0, o1, o2, o3, E, C, P, d1b);
+    else { g_err = "dm_mace_tp: C must be 64 or 128"; return -1; }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_mace_tp_bwd_f32(const float* g0, const float* g1, const float* g2,
+                       const float* g3, const float* x0, const float* x1,
+                       const float* Y, const float* w, const int32_t* nz,
+                       const float* nzc, int32_t nnz, float* dx0,
+                       float* dx1, float* dY, float* dw, int64_t E,
+                       int32_t C, int32_t P, int32_t d1b, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (C == 64)
+        k_mace_tp_bwd<1><<<nblocks(E, 4), BLOCK, 0, s>>>(
+            g0, g1, g2, g3, x0, x1, Y, w, nz, nzc, nnz, dx0, dx1, dY, dw,
+            E, C, P, d1b);
+    else if (C == 128)
+        k_mace_tp_bwd<2><<<nblocks(E, 4), BLOCK, 0, s>>>(
+            g0, g1, g2, g3, x0, x1, Y, w, nz, nzc, nnz, dx0, dx1, dY, dw,
+            E, C, P, d1b);
+    else { g_err = "dm_mace_tp: C must be 64 or 128"; return -1; }
     DM_CHECK_LAUNCH();
     return 0;
 }

@@ -308,3 +308,95 @@ def pad_missing(x: Dict[int, torch.Tensor], ls, like: torch.Tensor
         if l not in out:
             out[l] = like.new_zeros(N, C, 2 * l + 1)
     return out
+
+
+# ---------------------------------------------------------------------------
+# fused HIP tensor product (round 2): one wave per edge, CG nonzeros via
+# the scalar cache, LDS accumulators — include/distmlip_hip.h
+# ---------------------------------------------------------------------------
+
+@lru_cache(maxsize=None)
+def _tp_nz(paths_key, device_str):
+    import numpy as np
+    ents, coefs = [], []
+    for p, (l1, l2, l3) in enumerate(paths_key):
+        CG = so3.real_cg(l1, l2, l3)
+        slot = 0 if l1 == 0 else 1
+        for k1 in range(CG.shape[0]):
+            for k2l in range(CG.shape[1]):
+                for k3l in range(CG.shape[2]):
+                    c = CG[k1, k2l, k3l]
+                    if abs(c) > 1e-12:
+                        ents.append([p, slot, k1,
+                                     so3.L_OFF[l2] + k2l,
+                                     so3.L_OFF[l3] + k3l])
+                        coefs.append(c)
+    dev = torch.device(device_str)
+    nz = torch.tensor(np.asarray(ents, dtype=np.int32),
+                      device=dev).contiguous().view(-1)
+    nzc = torch.tensor(np.asarray(coefs, dtype=np.float32), device=dev)
+    return nz, nzc
+
+
+class _MaceTPFn(torch.autograd.Function):
+    """Fused uvu TP: m [E,16,C] from x0 [E,C], x1 [E,C,d1b] (or None),
+    Y [E,16], w [E,P,C] — forward and backward each ONE kernel."""
+
+    @staticmethod
+    def forward(ctx, x0, x1, Y, w, nz, nzc, P, d1b):
+        from distmlip_amd.ops import _check, _fp, _ip, _stream, hip_lib
+        E, C = x0.shape
+        outs = [torch.empty(E, 2 * l3 + 1, C, dtype=x0.dtype,
+                            device=x0.device) for l3 in range(4)]
+        _check(hip_lib().dm_mace_tp_fwd_f32(
+            _fp(x0), _fp(x1) if x1 is not None else None, _fp(Y), _fp(w),
+            _ip(nz), _fp(nzc), nz.numel() // 5,
+            _fp(outs[0]), _fp(outs[1]), _fp(outs[2]), _fp(outs[3]),
+            E, C, P, max(d1b, 1), _stream()), "dm_mace_tp_fwd_f32")
+        ctx.save_for_backward(x0, Y, w, nz, nzc)
+        ctx.x1 = x1
+        ctx.meta = (P, d1b)
+        return tuple(outs)
+
+    @staticmethod
+    def backward(ctx, g0, g1, g2, g3):
+        from distmlip_amd.ops import _check, _fp, _ip, _stream, hip_lib
+        x0, Y, w, nz, nzc = ctx.saved_tensors
+        x1 = ctx.x1
+        P, d1b = ctx.meta
+        E, C = x0.shape
+        gs = [g.contiguous() for g in (g0, g1, g2, g3)]
+        dx0 = torch.empty_like(x0)
+        dx1 = torch.empty_like(x1) if x1 is not None else None
+        dY = torch.empty_like(Y)
+        dw = torch.empty_like(w)
+        _check(hip_lib().dm_mace_tp_bwd_f32(
+            _fp(gs[0]), _fp(gs[1]), _fp(gs[2]), _fp(gs[3]),
+            _fp(x0), _fp(x1) if x1 is not None else None, _fp(Y),
+            _fp(w), _ip(nz), _fp(nzc), nz.numel() // 5, _fp(dx0),
+            _fp(dx1) if dx1 is not None else None, _fp(dY), _fp(dw),
+            E, C, P, max(d1b, 1), _stream()), "dm_mace_tp_bwd_f32")
+        return dx0, dx1, dY, dw, None, None, None, None
+
+
+def conv_tp_hip_available(inter: Interaction, Y: torch.Tensor,
+                          C: int) -> bool:
+    import os
+    return (os.environ.get("DM_MACE_TP", "hip") == "hip"
+            and Y.is_cuda and C in (64, 128)
+            and all(l1 in (0, 1) for (l1, _, _) in inter.paths))
+
+
+def conv_tp_hip(inter: Interaction, x0: torch.Tensor,
+                x1, Y: torch.Tensor, tp_w: torch.Tensor):
+    """Returns (m0..m3): per-l3 messages [E, 2*l3+1, C] (contiguous rows
+    for the scatter; note the l-major layout vs the torch path's
+    [E, C, d3])."""
+    paths_key = tuple(inter.paths)
+    nz, nzc = _tp_nz(paths_key, str(Y.device))
+    d1b = x1.shape[-1] if x1 is not None else 0
+    P = len(inter.paths)
+    return _MaceTPFn.apply(x0.contiguous(),
+                           x1.contiguous() if x1 is not None else None,
+                           Y.contiguous(), tp_w.contiguous(), nz, nzc,
+                           P, d1b)

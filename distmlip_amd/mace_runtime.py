@@ -181,10 +181,27 @@ class MaceSpmdEngine:
                     l: ops.gather(_flat(x_up[l]).contiguous(), pd.src,
                                   csr=src_csr).view(-1, C, 2 * l + 1)
                     for l in x_up}
-                msgs = mace_ops.conv_tp_messages(_inter, gathered, Y, tp_w)
-                m = {l3: ops.scatter_edges(
-                        _flat(msgs[l3]).contiguous(), pd
-                     ).view(-1, C, 2 * l3 + 1) for l3 in msgs}
+                if mace_ops.conv_tp_hip_available(_inter, Y, C):
+                    # fused per-edge TP kernel (one wave per edge,
+                    # include/distmlip_hip.h); outputs are [E, d3, C]
+                    # l-major — transpose back after the scatter ([N,*]
+                    # rows, cheap)
+                    mts = mace_ops.conv_tp_hip(
+                        _inter, gathered[0][:, :, 0],
+                        gathered.get(1), Y, tp_w)
+                    m = {}
+                    for l3 in range(4):
+                        d3 = 2 * l3 + 1
+                        sc = ops.scatter_edges(
+                            _flat(mts[l3]).contiguous(), pd
+                        ).view(-1, d3, C)
+                        m[l3] = sc.permute(0, 2, 1).contiguous()
+                else:
+                    msgs = mace_ops.conv_tp_messages(_inter, gathered, Y,
+                                                     tp_w)
+                    m = {l3: ops.scatter_edges(
+                            _flat(msgs[l3]).contiguous(), pd
+                         ).view(-1, C, 2 * l3 + 1) for l3 in msgs}
                 m = mace_ops.irreps_linear(_inter.linear_post, m)
                 m = {l: t / cfg.avg_num_neighbors for l, t in m.items()}
                 sc = mace_ops.skip_tp(_inter, xd, species)

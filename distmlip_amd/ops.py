@@ -57,6 +57,15 @@ def hip_lib() -> ctypes.CDLL:
         lib.dm_gated_combine_bwd_f32.restype = c_int32
         lib.dm_gated_combine_bwd_f32.argtypes = [fp, fp, fp, fp, fp, fp, fp,
                                                  c_int64, c_uint64]
+        lib.dm_mace_tp_fwd_f32.restype = c_int32
+        lib.dm_mace_tp_fwd_f32.argtypes = [fp, fp, fp, fp, ip, fp, c_int32,
+                                           fp, fp, fp, fp, c_int64, c_int32,
+                                           c_int32, c_int32, c_uint64]
+        lib.dm_mace_tp_bwd_f32.restype = c_int32
+        lib.dm_mace_tp_bwd_f32.argtypes = [fp, fp, fp, fp, fp, fp, fp, fp,
+                                           ip, fp, c_int32, fp, fp, fp, fp,
+                                           c_int64, c_int32, c_int32,
+                                           c_int32, c_uint64]
         cf = ctypes.c_float
         lib.dm_edge_geom_rbf_fwd_f32.restype = c_int32
         lib.dm_edge_geom_rbf_fwd_f32.argtypes = [fp, ip, ip, fp, fp, cf,

@@ -152,6 +152,27 @@ int dm_nl_fill_f64(const double* pos, const int32_t* cid,
                    const int32_t* row_ptr, int32_t* src, int8_t* off_i8,
                    uint8_t* bond_flag, int64_t N, uint64_t stream);
 
+/* MACE uvu tensor product, fused per edge (round 2).  Replaces the
+ * e3nn TensorProduct the reference's MACE interactions delegate to
+ * (implementations/mace/models.py:144-152, conv_tp of
+ * RealAgnosticResidualInteractionBlock): per-edge contraction of sender
+ * features (l=0 block x0 [E,C]; optional l=1 block x1 [E,C,d1b]) with
+ * edge spherical harmonics Y [E,16] and per-edge path weights w [E,P,C],
+ * CG nonzeros streamed as (path, slot, k1, k2, k3) int32 quintuples nz
+ * sorted by path + coefficients nzc.  m is [E,16,C] (l3-major rows).
+ * Backward writes dx0/dx1/dY/dw in the matching layouts. */
+int dm_mace_tp_fwd_f32(const float* x0, const float* x1, const float* Y,
+                       const float* w, const int32_t* nz, const float* nzc,
+                       int32_t nnz, float* o0, float* o1, float* o2,
+                       float* o3, int64_t E, int32_t C, int32_t P,
+                       int32_t d1b, uint64_t stream);
+int dm_mace_tp_bwd_f32(const float* g0, const float* g1, const float* g2,
+                       const float* g3, const float* x0, const float* x1,
+                       const float* Y, const float* w, const int32_t* nz,
+                       const float* nzc, int32_t nnz, float* dx0,
+                       float* dx1, float* dY, float* dw, int64_t E,
+                       int32_t C, int32_t P, int32_t d1b, uint64_t stream);
+
 const char* dm_hip_last_error(void);
 
 #ifdef __cplusplus
