@@ -810,6 +810,136 @@ __global__ void k_mace_tp_bwd(const float* __restrict__ g0,
 }
 
 
+
+// ---------------------------------------------------------------------------
+// MACE symmetric contraction, fused per node (round 2).  Replaces the
+// SymmetricContraction of mace's EquivariantProductBasisBlock (the
+// reference delegates to mace.modules.symmetric_contraction via
+// models.py:155-157): out[n, orow, c] = sum_entries
+// W[elem(n), wrow, c] * coef * x[n,a,c] * x[n,b,c] * x[n,k,c], with the
+// sentinel row x[16] == 1 encoding nu<3 terms.  Entries are wave-uniform
+// (scalar cache), sorted by wrow so each weight row loads once; weights
+// are FROZEN (inference engine), so backward produces dx only.
+// ---------------------------------------------------------------------------
+
+template <int CPL>
+__global__ void k_mace_symc_fwd(const float* __restrict__ x,
+                                const int32_t* __restrict__ elem,
+                                const float* __restrict__ W,
+                                const int32_t* __restrict__ nz,
+                                const float* __restrict__ nzc,
+                                int32_t nnz, float* __restrict__ out,
+                                int64_t N, int32_t C, int32_t T,
+                                int32_t S_out) {
+    // x rows AND accumulators in LDS: a/b/k/orow are runtime indices
+    // (dynamic register-array indexing would spill to scratch)
+    __shared__ float acc[4 * 4 * 64 * CPL];       // [orow][wave][lane][i]
+    __shared__ float xls[17 * 4 * 64 * CPL];      // [row][wave][lane][i]
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+#define XR(a, i) xls[(((a) * 4 + wid) * 64 + lane) * CPL + (i)]
+    for (int64_t n = blockIdx.x * (int64_t)nw + wid; n < N;
+         n += (int64_t)gridDim.x * nw) {
+        for (int a = 0; a < 16; ++a)
+            for (int i = 0; i < CPL; ++i)
+                XR(a, i) = x[((int64_t)n * 16 + a) * C + lane + i * 64];
+        for (int i = 0; i < CPL; ++i)
+            XR(16, i) = 1.0f;
+        for (int o = 0; o < S_out; ++o)
+            for (int i = 0; i < CPL; ++i)
+                acc[((o * 4 + wid) * 64 + lane) * CPL + i] = 0.0f;
+        const int64_t wbase = (int64_t)elem[n] * T;
+        int curw = -1;
+        float wreg[CPL];
+        for (int32_t t = 0; t < nnz; ++t) {
+            const int32_t wrow = nz[6 * t], a = nz[6 * t + 1],
+                          b = nz[6 * t + 2], k = nz[6 * t + 3],
+                          orow = nz[6 * t + 4];
+            if (wrow != curw) {
+                curw = wrow;
+                for (int i = 0; i < CPL; ++i)
+                    wreg[i] = W[(wbase + wrow) * C + lane + i * 64];
+            }
+            const float c = nzc[t];
+            for (int i = 0; i < CPL; ++i) {
+                const float v = c * wreg[i] * XR(a, i) * XR(b, i)
+                                * XR(k, i);
+                acc[((orow * 4 + wid) * 64 + lane) * CPL + i] += v;
+            }
+        }
+        for (int o = 0; o < S_out; ++o)
+            for (int i = 0; i < CPL; ++i)
+                out[((int64_t)n * S_out + o) * C + lane + i * 64] =
+                    acc[((o * 4 + wid) * 64 + lane) * CPL + i];
+    }
+#undef XR
+}
+
+template <int CPL>
+__global__ void k_mace_symc_bwd(const float* __restrict__ go,
+                                const float* __restrict__ x,
+                                const int32_t* __restrict__ elem,
+                                const float* __restrict__ W,
+                                const int32_t* __restrict__ nz,
+                                const float* __restrict__ nzc,
+                                int32_t nnz, float* __restrict__ dx,
+                                int64_t N, int32_t C, int32_t T,
+                                int32_t S_out) {
+    __shared__ float dacc[17 * 4 * 64 * CPL];     // row 16 = sentinel sink
+    __shared__ float xls[17 * 4 * 64 * CPL];
+    __shared__ float gls[4 * 4 * 64 * CPL];
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+#define XRB(a, i) xls[(((a) * 4 + wid) * 64 + lane) * CPL + (i)]
+#define GOB(o, i) gls[(((o) * 4 + wid) * 64 + lane) * CPL + (i)]
+    for (int64_t n = blockIdx.x * (int64_t)nw + wid; n < N;
+         n += (int64_t)gridDim.x * nw) {
+        for (int a = 0; a < 16; ++a)
+            for (int i = 0; i < CPL; ++i)
+                XRB(a, i) = x[((int64_t)n * 16 + a) * C + lane + i * 64];
+        for (int i = 0; i < CPL; ++i)
+            XRB(16, i) = 1.0f;
+        for (int o = 0; o < S_out; ++o)
+            for (int i = 0; i < CPL; ++i)
+                GOB(o, i) = go[((int64_t)n * S_out + o) * C + lane
+                               + i * 64];
+        for (int a = 0; a < 17; ++a)
+            for (int i = 0; i < CPL; ++i)
+                dacc[((a * 4 + wid) * 64 + lane) * CPL + i] = 0.0f;
+        const int64_t wbase = (int64_t)elem[n] * T;
+        int curw = -1;
+        float wreg[CPL];
+        for (int32_t t = 0; t < nnz; ++t) {
+            const int32_t wrow = nz[6 * t], a = nz[6 * t + 1],
+                          b = nz[6 * t + 2], k = nz[6 * t + 3],
+                          orow = nz[6 * t + 4];
+            if (wrow != curw) {
+                curw = wrow;
+                for (int i = 0; i < CPL; ++i)
+                    wreg[i] = W[(wbase + wrow) * C + lane + i * 64];
+            }
+            const float c = nzc[t];
+            for (int i = 0; i < CPL; ++i) {
+                const float g = c * wreg[i] * GOB(orow, i);
+                const float xa = XRB(a, i), xb = XRB(b, i),
+                            xk = XRB(k, i);
+                dacc[((a * 4 + wid) * 64 + lane) * CPL + i] += g * xb * xk;
+                dacc[((b * 4 + wid) * 64 + lane) * CPL + i] += g * xa * xk;
+                dacc[((k * 4 + wid) * 64 + lane) * CPL + i] += g * xa * xb;
+            }
+        }
+        for (int a = 0; a < 16; ++a)
+            for (int i = 0; i < CPL; ++i)
+                dx[((int64_t)n * 16 + a) * C + lane + i * 64] =
+                    dacc[((a * 4 + wid) * 64 + lane) * CPL + i];
+    }
+#undef XRB
+#undef GOB
+}
+
+
 // ---------------------------------------------------------------------------
 // C ABI
 // ---------------------------------------------------------------------------
@@ -1083,6 +1213,43 @@ int dm_mace_tp_bwd_f32(const float* g0, const float* g1, const float* g2,
             g0, g1, g2, g3, x0, x1, Y, w, nz, nzc, nnz, dx0, dx1, dY, dw,
             E, C, P, d1b);
     else { g_err = "dm_mace_tp: C must be 64 or 128"; return -1; }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+
+int dm_mace_symc_fwd_f32(const float* x, const int32_t* elem,
+                         const float* W, const int32_t* nz,
+                         const float* nzc, int32_t nnz, float* out,
+                         int64_t N, int32_t C, int32_t T, int32_t S_out,
+                         uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (S_out > 4) { g_err = "dm_mace_symc: S_out > 4"; return -1; }
+    if (C == 64)
+        k_mace_symc_fwd<1><<<nblocks(N, 4), BLOCK, 0, s>>>(
+            x, elem, W, nz, nzc, nnz, out, N, C, T, S_out);
+    else if (C == 128)
+        k_mace_symc_fwd<2><<<nblocks(N, 4), BLOCK, 0, s>>>(
+            x, elem, W, nz, nzc, nnz, out, N, C, T, S_out);
+    else { g_err = "dm_mace_symc: C must be 64 or 128"; return -1; }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_mace_symc_bwd_f32(const float* go, const float* x,
+                         const int32_t* elem, const float* W,
+                         const int32_t* nz, const float* nzc, int32_t nnz,
+                         float* dx, int64_t N, int32_t C, int32_t T,
+                         int32_t S_out, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (S_out > 4) { g_err = "dm_mace_symc: S_out > 4"; return -1; }
+    if (C == 64)
+        k_mace_symc_bwd<1><<<nblocks(N, 4), BLOCK, 0, s>>>(
+            go, x, elem, W, nz, nzc, nnz, dx, N, C, T, S_out);
+    else if (C == 128)
+        k_mace_symc_bwd<2><<<nblocks(N, 4), BLOCK, 0, s>>>(
+            go, x, elem, W, nz, nzc, nnz, dx, N, C, T, S_out);
+    else { g_err = "dm_mace_symc: C must be 64 or 128"; return -1; }
     DM_CHECK_LAUNCH();
     return 0;
 }

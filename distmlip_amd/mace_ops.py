@@ -262,6 +262,8 @@ def symmetric_contract(prod: ProductBasis, x: Dict[int, torch.Tensor],
     tiny tiles — the rocBLAS ~1%-of-peak shape, see
     conv_tp_messages.)"""
     like = next(iter(x.values()))
+    if symc_hip_available(x, prod):
+        return symmetric_contract_hip(prod, x, species, correlation)
     N, C = like.shape[0], like.shape[1]
     dts, devs = str(like.dtype).split(".")[-1], str(like.device)
     out = {}
@@ -400,3 +402,150 @@ def conv_tp_hip(inter: Interaction, x0: torch.Tensor,
                            x1.contiguous() if x1 is not None else None,
                            Y.contiguous(), tp_w.contiguous(), nz, nzc,
                            P, d1b)
+
+
+# ---------------------------------------------------------------------------
+# fused HIP symmetric contraction (round 2): one wave per node, polynomial
+# entries via the scalar cache, x rows + accumulators in LDS.  Weights are
+# FROZEN (inference engine) — backward produces dx only, so the
+# availability check requires requires_grad=False weights.
+# ---------------------------------------------------------------------------
+
+@lru_cache(maxsize=None)
+def _symc_nz(out_ls, correlation: int, device_str: str):
+    import numpy as np
+    ents, coefs = [], []
+    wrow = 0
+    o_off = 0
+    meta = []                      # (lo, nu, n_trees) in wrow order
+    for lo in out_ls:
+        do = 2 * lo + 1
+        for nu in range(1, correlation + 1):
+            trees, _M = so3.symmetric_basis_trees(nu, lo)
+            for tr in trees:
+                if nu == 1:
+                    (l,) = tr
+                    for m in range(do):
+                        ents.append([wrow, so3.L_OFF[l] + m, 16, 16,
+                                     o_off + m])
+                        coefs.append(1.0)
+                elif nu == 2:
+                    l1, l2 = tr
+                    C = so3.real_cg(l1, l2, lo)
+                    for a in range(C.shape[0]):
+                        for b in range(C.shape[1]):
+                            for oo in range(do):
+                                c = C[a, b, oo]
+                                if abs(c) > 1e-12:
+                                    ents.append([wrow,
+                                                 so3.L_OFF[l1] + a,
+                                                 so3.L_OFF[l2] + b, 16,
+                                                 o_off + oo])
+                                    coefs.append(c)
+                else:
+                    l1, l2, L, l3 = tr
+                    K = np.einsum("abL,Lko->abko",
+                                  so3.real_cg(l1, l2, L),
+                                  so3.real_cg(L, l3, lo))
+                    nzs = np.argwhere(np.abs(K) > 1e-12)
+                    for (a, b, k, oo) in nzs:
+                        ents.append([wrow, so3.L_OFF[l1] + a,
+                                     so3.L_OFF[l2] + b,
+                                     so3.L_OFF[l3] + k, o_off + oo])
+                        coefs.append(K[a, b, k, oo])
+                wrow += 1
+            meta.append((lo, nu, len(trees)))
+        o_off += do
+    dev = torch.device(device_str)
+    # kernel reads sextuples; pad with a zero column
+    arr = np.asarray(ents, dtype=np.int32)
+    arr6 = np.zeros((len(ents), 6), dtype=np.int32)
+    arr6[:, :5] = arr
+    nz = torch.tensor(arr6, device=dev).contiguous().view(-1)
+    nzc = torch.tensor(np.asarray(coefs, dtype=np.float32), device=dev)
+    return nz, nzc, wrow, o_off, tuple(meta)
+
+
+def _symc_wcat(prod: ProductBasis, correlation: int) -> torch.Tensor:
+    """Per-element weight table [n_elem, T_total, C] in wrow order
+    (tree-major within each (lo, nu)); cached on the module while
+    frozen."""
+    key = (next(iter(prod.weights.values())).device,
+           tuple(p._version for p in prod.weights.values()))
+    cached = getattr(prod, "_dm_wcat", None)
+    if cached is not None and cached[0] == key:
+        return cached[1]
+    dts = str(next(iter(prod.weights.values())).dtype).split(".")[-1]
+    devs = str(next(iter(prod.weights.values())).device)
+    cols = []
+    for lo in prod.out_ls:
+        for nu in range(1, correlation + 1):
+            k = f"{lo}_{nu}"
+            if k not in prod.weights:
+                continue
+            _, M = _trees_M(nu, lo, dts, devs)
+            w = prod.weights[k]                      # [n_elem, P, C]
+            cols.append(torch.einsum("epc,pt->etc", w, M))
+    W = torch.cat(cols, dim=1).contiguous()
+    if not any(p.requires_grad for p in prod.weights.values()):
+        prod._dm_wcat = (key, W)
+    return W
+
+
+class _MaceSymcFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, xs, elem, W, nz, nzc, S_out):
+        from distmlip_amd.ops import _check, _fp, _ip, _stream, hip_lib
+        N, _, C = xs.shape
+        T = W.shape[1]
+        out = torch.empty(N, S_out, C, dtype=xs.dtype, device=xs.device)
+        _check(hip_lib().dm_mace_symc_fwd_f32(
+            _fp(xs), _ip(elem), _fp(W), _ip(nz), _fp(nzc),
+            nz.numel() // 6, _fp(out), N, C, T, S_out, _stream()),
+            "dm_mace_symc_fwd_f32")
+        ctx.save_for_backward(xs, elem, W, nz, nzc)
+        ctx.S_out = S_out
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        from distmlip_amd.ops import _check, _fp, _ip, _stream, hip_lib
+        xs, elem, W, nz, nzc = ctx.saved_tensors
+        N, _, C = xs.shape
+        T = W.shape[1]
+        dx = torch.empty_like(xs)
+        _check(hip_lib().dm_mace_symc_bwd_f32(
+            _fp(go.contiguous()), _fp(xs), _ip(elem), _fp(W), _ip(nz),
+            _fp(nzc), nz.numel() // 6, _fp(dx), N, C, T, ctx.S_out,
+            _stream()), "dm_mace_symc_bwd_f32")
+        return dx, None, None, None, None, None
+
+
+def symmetric_contract_hip(prod: ProductBasis, x: Dict[int, torch.Tensor],
+                           species: torch.Tensor, correlation: int
+                           ) -> Dict[int, torch.Tensor]:
+    like = next(iter(x.values()))
+    N, C = like.shape[0], like.shape[1]
+    xs = torch.cat([x[l].permute(0, 2, 1) for l in sorted(x)],
+                   dim=1).contiguous()            # [N, 16, C]
+    nz, nzc, T, S_out, _ = _symc_nz(tuple(prod.out_ls), correlation,
+                                    str(like.device))
+    W = _symc_wcat(prod, correlation)
+    elem = species.to(torch.int32).contiguous()
+    out = _MaceSymcFn.apply(xs, elem, W, nz, nzc, S_out)
+    res, off = {}, 0
+    for lo in prod.out_ls:
+        do = 2 * lo + 1
+        res[lo] = out[:, off:off + do, :].permute(0, 2, 1).contiguous()
+        off += do
+    return res
+
+
+def symc_hip_available(x: Dict[int, torch.Tensor],
+                       prod: ProductBasis) -> bool:
+    import os
+    like = next(iter(x.values()))
+    return (os.environ.get("DM_MACE_SYMC", "hip") == "hip"
+            and like.is_cuda and like.shape[1] in (64, 128)
+            and sorted(x) == [0, 1, 2, 3]
+            and not any(p.requires_grad for p in prod.weights.values()))

@@ -173,6 +173,23 @@ int dm_mace_tp_bwd_f32(const float* g0, const float* g1, const float* g2,
                        float* dx1, float* dY, float* dw, int64_t E,
                        int32_t C, int32_t P, int32_t d1b, uint64_t stream);
 
+/* MACE symmetric contraction, fused per node (round 2).  Replaces
+ * mace's SymmetricContraction (EquivariantProductBasisBlock,
+ * models.py:155-157): polynomial entries (wrow, a, b, k, orow) int32
+ * sextuples + coefficients, x rows [N,16,C] (sentinel row 16 == 1
+ * implied), per-element weight table W [n_elem, T, C]; weights frozen
+ * so backward produces dx only. */
+int dm_mace_symc_fwd_f32(const float* x, const int32_t* elem,
+                         const float* W, const int32_t* nz,
+                         const float* nzc, int32_t nnz, float* out,
+                         int64_t N, int32_t C, int32_t T, int32_t S_out,
+                         uint64_t stream);
+int dm_mace_symc_bwd_f32(const float* go, const float* x,
+                         const int32_t* elem, const float* W,
+                         const int32_t* nz, const float* nzc, int32_t nnz,
+                         float* dx, int64_t N, int32_t C, int32_t T,
+                         int32_t S_out, uint64_t stream);
+
 const char* dm_hip_last_error(void);
 
 #ifdef __cplusplus

@@ -124,3 +124,49 @@ def test_fused_tp_kernel_vs_torch():
         assert torch.allclose(wk.grad, wr.grad, atol=2e-4), li
         dY_err = (Yk.grad - Yr.grad).abs().max().item()
         assert dY_err < 2e-3, (li, dY_err)   # dY sums over C=128 lanes
+
+
+@requires_gpu
+def test_fused_symc_kernel_vs_torch():
+    """dm_mace_symc_{fwd,bwd}_f32 against the torch combo-GEMM
+    contraction: values + dx gradients, both product shapes (hidden
+    0e+1o; scalar-only last layer)."""
+    import os
+
+    from distmlip_amd import mace_ops
+    from distmlip_amd.mace_model import MACEConfig, MACECore
+
+    torch.manual_seed(1)
+    dev = torch.device("cuda:0")
+    cfg = MACEConfig(n_elements=3, channels=128)
+    core = MACECore.seeded(cfg, seed=0).float().to(dev)
+    core.requires_grad_(False)
+    N = 3000
+    species = torch.randint(0, 3, (N,), device=dev)
+    for li in (0, 1):
+        prod = core.products[li]
+        x = {l: torch.randn(N, cfg.channels, 2 * l + 1, device=dev)
+             for l in range(4)}
+        xr = {l: t.clone().requires_grad_(True) for l, t in x.items()}
+        os.environ["DM_MACE_SYMC"] = "torch"
+        try:
+            ref = mace_ops.symmetric_contract(prod, xr, species,
+                                              cfg.correlation)
+        finally:
+            os.environ.pop("DM_MACE_SYMC", None)
+        lw = {lo: torch.randn_like(ref[lo]) for lo in ref}
+        sum((ref[lo] * lw[lo]).sum() for lo in ref).backward()
+
+        xk = {l: t.clone().requires_grad_(True) for l, t in x.items()}
+        got = mace_ops.symmetric_contract_hip(prod, xk, species,
+                                              cfg.correlation)
+        loss = 0
+        for lo in ref:
+            err = (got[lo] - ref[lo]).abs().max().item()
+            assert err < 5e-4, (li, lo, err)
+            loss = loss + (got[lo] * lw[lo]).sum()
+        loss.backward()
+        for l in range(4):
+            derr = (xk[l].grad - xr[l].grad).abs().max().item()
+            scale = xr[l].grad.abs().max().item()
+            assert derr < 1e-3 * max(1.0, scale), (li, l, derr, scale)
