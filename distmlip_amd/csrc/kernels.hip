@@ -940,6 +940,118 @@ __global__ void k_mace_symc_bwd(const float* __restrict__ go,
 }
 
 
+
+// ---------------------------------------------------------------------------
+// eSCN/UMA per-edge Wigner rotations (round 2).  The reference's UMA path
+// applies per-edge block-diagonal Wigner matrices around every SO(2)
+// convolution (escn_md.py:283-291, delegated to fairchem/e3nn); as batched
+// bmms of (S x S)(S x C) tiles these run rocBLAS at ~1% of peak (62% of a
+// 16 s step, profiles/r2_uma_kernel_stats.csv).  Here: one wave per edge
+// (gather/apply) or per node (apply/scatter over the dst CSR), D rows
+// streamed wave-uniform through the scalar cache, channels across lanes.
+// S = 9 (lmax 2).  trans selects D vs D^T (the inverse rotation).
+// ---------------------------------------------------------------------------
+
+template <int CPL>
+__global__ void k_rot_gather(const float* __restrict__ h,
+                             const int32_t* __restrict__ idx,
+                             const float* __restrict__ D, int32_t trans,
+                             float* __restrict__ out, int64_t E,
+                             int32_t C) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+    for (int64_t e = blockIdx.x * (int64_t)nw + wid; e < E;
+         e += (int64_t)gridDim.x * nw) {
+        const int64_t row = idx ? (int64_t)idx[e] : e;
+        float hr[9][CPL];
+        for (int t = 0; t < 9; ++t)
+            for (int i = 0; i < CPL; ++i)
+                hr[t][i] = h[(row * 9 + t) * C + lane + i * 64];
+        const float* De = D + e * 81;
+        for (int s = 0; s < 9; ++s) {
+            float acc[CPL] = {};
+            for (int t = 0; t < 9; ++t) {
+                const float d = trans ? De[t * 9 + s] : De[s * 9 + t];
+                for (int i = 0; i < CPL; ++i)
+                    acc[i] += d * hr[t][i];
+            }
+            for (int i = 0; i < CPL; ++i)
+                out[((int64_t)e * 9 + s) * C + lane + i * 64] = acc[i];
+        }
+    }
+}
+
+template <int CPL>
+__global__ void k_rot_scatter(const float* __restrict__ mt,
+                              const float* __restrict__ D, int32_t trans,
+                              const int32_t* __restrict__ perm,
+                              const int32_t* __restrict__ row_ptr,
+                              const float* __restrict__ base,
+                              float* __restrict__ out, int64_t N,
+                              int32_t C) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+    for (int64_t n = blockIdx.x * (int64_t)nw + wid; n < N;
+         n += (int64_t)gridDim.x * nw) {
+        float acc[9][CPL] = {};
+        const int32_t lo = row_ptr[n], hi = row_ptr[n + 1];
+        for (int32_t q = lo; q < hi; ++q) {
+            const int64_t e = perm ? (int64_t)perm[q] : (int64_t)q;
+            const float* De = D + e * 81;
+            float mr[9][CPL];
+            for (int t = 0; t < 9; ++t)
+                for (int i = 0; i < CPL; ++i)
+                    mr[t][i] = mt[(e * 9 + t) * C + lane + i * 64];
+            for (int s = 0; s < 9; ++s)
+                for (int t = 0; t < 9; ++t) {
+                    const float d = trans ? De[t * 9 + s] : De[s * 9 + t];
+                    for (int i = 0; i < CPL; ++i)
+                        acc[s][i] += d * mr[t][i];
+                }
+        }
+        for (int s = 0; s < 9; ++s)
+            for (int i = 0; i < CPL; ++i) {
+                const int64_t o = ((int64_t)n * 9 + s) * C + lane + i * 64;
+                out[o] = acc[s][i] + (base ? base[o] : 0.0f);
+            }
+    }
+}
+
+template <int CPL>
+__global__ void k_rot_dD(const float* __restrict__ go,
+                         const float* __restrict__ h,
+                         const int32_t* __restrict__ idx, int32_t trans,
+                         float* __restrict__ dD, int64_t E, int32_t C) {
+    // dD(s,t) = sum_c go[e,s,c] * in[row,t,c]; trans writes transposed
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+    for (int64_t e = blockIdx.x * (int64_t)nw + wid; e < E;
+         e += (int64_t)gridDim.x * nw) {
+        const int64_t row = idx ? (int64_t)idx[e] : e;
+        float hr[9][CPL], gr[9][CPL];
+        for (int t = 0; t < 9; ++t)
+            for (int i = 0; i < CPL; ++i) {
+                hr[t][i] = h[(row * 9 + t) * C + lane + i * 64];
+                gr[t][i] = go[((int64_t)e * 9 + t) * C + lane + i * 64];
+            }
+        for (int s = 0; s < 9; ++s)
+            for (int t = 0; t < 9; ++t) {
+                float p = 0.0f;
+                for (int i = 0; i < CPL; ++i)
+                    p += gr[s][i] * hr[t][i];
+                // wave reduction (butterfly over 64 lanes)
+                for (int off = 32; off > 0; off >>= 1)
+                    p += __shfl_xor(p, off, 64);
+                if (lane == 0)
+                    dD[e * 81 + (trans ? t * 9 + s : s * 9 + t)] = p;
+            }
+    }
+}
+
+
 // ---------------------------------------------------------------------------
 // C ABI
 // ---------------------------------------------------------------------------
@@ -1250,6 +1362,53 @@ int dm_mace_symc_bwd_f32(const float* go, const float* x,
         k_mace_symc_bwd<2><<<nblocks(N, 4), BLOCK, 0, s>>>(
             go, x, elem, W, nz, nzc, nnz, dx, N, C, T, S_out);
     else { g_err = "dm_mace_symc: C must be 64 or 128"; return -1; }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+
+int dm_rot_gather_f32(const float* h, const int32_t* idx, const float* D,
+                      int32_t trans, float* out, int64_t E, int32_t C,
+                      uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (C == 64)
+        k_rot_gather<1><<<nblocks(E, 4), BLOCK, 0, s>>>(h, idx, D, trans,
+                                                        out, E, C);
+    else if (C == 128)
+        k_rot_gather<2><<<nblocks(E, 4), BLOCK, 0, s>>>(h, idx, D, trans,
+                                                        out, E, C);
+    else { g_err = "dm_rot: C must be 64 or 128"; return -1; }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_rot_scatter_f32(const float* mt, const float* D, int32_t trans,
+                       const int32_t* perm, const int32_t* row_ptr,
+                       const float* base, float* out, int64_t N,
+                       int32_t C, uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (C == 64)
+        k_rot_scatter<1><<<nblocks(N, 4), BLOCK, 0, s>>>(
+            mt, D, trans, perm, row_ptr, base, out, N, C);
+    else if (C == 128)
+        k_rot_scatter<2><<<nblocks(N, 4), BLOCK, 0, s>>>(
+            mt, D, trans, perm, row_ptr, base, out, N, C);
+    else { g_err = "dm_rot: C must be 64 or 128"; return -1; }
+    DM_CHECK_LAUNCH();
+    return 0;
+}
+
+int dm_rot_dD_f32(const float* go, const float* h, const int32_t* idx,
+                  int32_t trans, float* dD, int64_t E, int32_t C,
+                  uint64_t stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (C == 64)
+        k_rot_dD<1><<<nblocks(E, 4), BLOCK, 0, s>>>(go, h, idx, trans, dD,
+                                                    E, C);
+    else if (C == 128)
+        k_rot_dD<2><<<nblocks(E, 4), BLOCK, 0, s>>>(go, h, idx, trans, dD,
+                                                    E, C);
+    else { g_err = "dm_rot: C must be 64 or 128"; return -1; }
     DM_CHECK_LAUNCH();
     return 0;
 }

@@ -61,6 +61,15 @@ def hip_lib() -> ctypes.CDLL:
         lib.dm_mace_tp_fwd_f32.argtypes = [fp, fp, fp, fp, ip, fp, c_int32,
                                            fp, fp, fp, fp, c_int64, c_int32,
                                            c_int32, c_int32, c_uint64]
+        lib.dm_rot_gather_f32.restype = c_int32
+        lib.dm_rot_gather_f32.argtypes = [fp, ip, fp, c_int32, fp,
+                                          c_int64, c_int32, c_uint64]
+        lib.dm_rot_scatter_f32.restype = c_int32
+        lib.dm_rot_scatter_f32.argtypes = [fp, fp, c_int32, ip, ip, fp,
+                                           fp, c_int64, c_int32, c_uint64]
+        lib.dm_rot_dD_f32.restype = c_int32
+        lib.dm_rot_dD_f32.argtypes = [fp, fp, ip, c_int32, fp, c_int64,
+                                      c_int32, c_uint64]
         lib.dm_mace_symc_fwd_f32.restype = c_int32
         lib.dm_mace_symc_fwd_f32.argtypes = [fp, ip, fp, ip, fp, c_int32,
                                              fp, c_int64, c_int32, c_int32,

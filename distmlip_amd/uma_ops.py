@@ -6,6 +6,7 @@ Primitives the UMA runtime composes per partition; the oracle
 from __future__ import annotations
 
 import math
+from functools import lru_cache
 from typing import Dict
 
 import torch
@@ -138,3 +139,143 @@ def energy_head(core: UMACore, x):
     s = h[:, 0, :]
     s = torch.nn.functional.silu(s @ core.head1.t())
     return core.scale * (s @ core.head2) + core.shift
+
+
+# ---------------------------------------------------------------------------
+# fused HIP rotation family (round 2): dm_rot_{gather,scatter,dD}_f32 —
+# replaces the per-edge Wigner bmms (62% of the uma250k step as rocBLAS
+# tiny-tile batched GEMMs, profiles/r2_uma_kernel_stats.csv)
+# ---------------------------------------------------------------------------
+
+def _rot_gather_raw(h, idx, D, trans):
+    from distmlip_amd.ops import _check, _fp, _ip, _stream, hip_lib
+    E = D.shape[0]
+    C = h.shape[-1]
+    out = torch.empty(E, 9, C, dtype=h.dtype, device=h.device)
+    _check(hip_lib().dm_rot_gather_f32(
+        _fp(h), _ip(idx) if idx is not None else None, _fp(D), trans,
+        _fp(out), E, C, _stream()), "dm_rot_gather_f32")
+    return out
+
+
+def _rot_dD_raw(go_e, h, idx, trans):
+    from distmlip_amd.ops import _check, _fp, _ip, _stream, hip_lib
+    E = go_e.shape[0]
+    C = go_e.shape[-1]
+    dD = torch.empty(E, 9, 9, dtype=go_e.dtype, device=go_e.device)
+    _check(hip_lib().dm_rot_dD_f32(
+        _fp(go_e), _fp(h), _ip(idx) if idx is not None else None, trans,
+        _fp(dD.view(E, 81)), E, C, _stream()), "dm_rot_dD_f32")
+    return dD
+
+
+class _RotGather(torch.autograd.Function):
+    """out[e] = D_e . h[idx[e]]   (h [N,9,C] f32, D [E,9,9] f32)."""
+
+    @staticmethod
+    def forward(ctx, h, D, idx):
+        h = h.contiguous()
+        D = D.contiguous()
+        out = _rot_gather_raw(h, idx, D.view(-1, 81), 0)
+        ctx.save_for_backward(h, D, idx)
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        h, D, idx = ctx.saved_tensors
+        go = go.contiguous()
+        # dh: scatter of D^T . go to the gathered rows
+        tmp = _rot_gather_raw(go, None, D.view(-1, 81), 1)
+        dh = torch.zeros_like(h).index_add_(0, idx.long(), tmp)
+        dD = _rot_dD_raw(go, h, idx, 0)
+        return dh, dD, None
+
+
+class _RotScatter(torch.autograd.Function):
+    """out[n] = sum_{e in row n} D_e^T . mt[e]  over a contiguous
+    (dst-sorted) row_ptr covering rows [0, N_local)."""
+
+    @staticmethod
+    def forward(ctx, mt, D, row_ptr, dst_local):
+        from distmlip_amd.ops import _check, _fp, _ip, _stream, hip_lib
+        mt = mt.contiguous()
+        D = D.contiguous()
+        N = row_ptr.shape[0] - 1
+        C = mt.shape[-1]
+        out = torch.empty(N, 9, C, dtype=mt.dtype, device=mt.device)
+        _check(hip_lib().dm_rot_scatter_f32(
+            _fp(mt), _fp(D.view(-1, 81)), 1, None, _ip(row_ptr), None,
+            _fp(out), N, C, _stream()), "dm_rot_scatter_f32")
+        ctx.save_for_backward(mt, D, dst_local)
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        mt, D, dst_local = ctx.saved_tensors
+        go = go.contiguous()
+        dmt = _rot_gather_raw(go, dst_local, D.view(-1, 81), 0)
+        dD = _rot_dD_raw(mt, go, dst_local, 0)
+        return dmt, dD, None, None
+
+
+def _so2_split(conv: SO2Conv):
+    """Per-side (src/dst) weight splits of the concatenated SO(2) mixes
+    (the cat over the channel dim becomes two tall GEMMs); cached on the
+    module while frozen."""
+    key = tuple(p._version for p in conv.parameters())
+    cached = getattr(conv, "_dm_split", None)
+    if cached is not None and cached[0] == key:
+        return cached[1]
+    lmax = conv.lmax
+    n0 = lmax + 1
+    C = conv.w0.shape[0] // n0
+    def split(w, nl):
+        v = w.view(w.shape[0], nl, 2, C)
+        return (v[:, :, 0, :].reshape(w.shape[0], nl * C).contiguous(),
+                v[:, :, 1, :].reshape(w.shape[0], nl * C).contiguous())
+    packs = {"w0": split(conv.w0, n0)}
+    for m in range(1, lmax + 1):
+        nl = lmax + 1 - m
+        packs[f"wr{m}"] = split(conv.wr[m - 1], nl)
+        packs[f"wi{m}"] = split(conv.wi[m - 1], nl)
+    if not any(p.requires_grad for p in conv.parameters()):
+        conv._dm_split = (key, packs)
+    return packs
+
+
+def so2_conv_split(conv: SO2Conv, xs: torch.Tensor, xd: torch.Tensor,
+                   gate: torch.Tensor, C: int) -> torch.Tensor:
+    """so2_conv over separate rotated src/dst tensors [E,9,C] (no cat
+    materialization; the concatenated GEMM splits exactly)."""
+    E = xs.shape[0]
+    lmax = conv.lmax
+    m0, pm = m_indices(lmax)
+    sp = _so2_split(conv)
+    out = xs.new_zeros(E, (lmax + 1) ** 2, C)
+    ws, wd = sp["w0"]
+    a0 = xs[:, m0, :].reshape(E, -1) @ ws.t() \
+        + xd[:, m0, :].reshape(E, -1) @ wd.t()
+    out[:, m0, :] = a0.view(E, lmax + 1, C) * gate[:, 0].view(E, 1, 1)
+    for m in range(1, lmax + 1):
+        plus, minus = pm[m - 1]
+        xps = xs[:, plus, :].reshape(E, -1)
+        xpd = xd[:, plus, :].reshape(E, -1)
+        xms = xs[:, minus, :].reshape(E, -1)
+        xmd = xd[:, minus, :].reshape(E, -1)
+        wrs, wrd = sp[f"wr{m}"]
+        wis, wid_ = sp[f"wi{m}"]
+        op = (xps @ wrs.t() + xpd @ wrd.t()
+              - xms @ wis.t() - xmd @ wid_.t())
+        om = (xps @ wis.t() + xpd @ wid_.t()
+              + xms @ wrs.t() + xmd @ wrd.t())
+        g = gate[:, m].view(E, 1, 1)
+        out[:, plus, :] = op.view(E, len(plus), C) * g
+        out[:, minus, :] = om.view(E, len(plus), C) * g
+    return out
+
+
+def rot_kernels_available(x_like: torch.Tensor, lmax: int) -> bool:
+    import os
+    return (os.environ.get("DM_UMA_ROT", "hip") == "hip"
+            and x_like.is_cuda and lmax == 2
+            and x_like.shape[-1] in (64, 128))

@@ -187,20 +187,56 @@ class UmaSpmdEngine:
             # chunks (dst-sorted edges), bounding transients
             chunk_edges = int(_os.environ.get("DM_UMA_CHUNK", 1_500_000))
             E_tot = len(pd.src)
+            N_loc = x.shape[0]
             if E_tot > chunk_edges:
                 rp = pd.row_ptr.long().cpu().numpy()
-                ranges = []
+                ranges = []            # (n0, n1, e0, e1), node-aligned
                 n0 = 0
                 for n in range(1, len(rp)):
                     if rp[n] - rp[n0] >= chunk_edges or n == len(rp) - 1:
-                        ranges.append((int(rp[n0]), int(rp[n])))
+                        ranges.append((n0, n, int(rp[n0]), int(rp[n])))
                         n0 = n
             else:
-                ranges = [(0, E_tot)]
+                ranges = [(0, N_loc, 0, E_tot)]
+            use_rot = uma_ops.rot_kernels_available(x, cfg.lmax) \
+                and not getattr(self.ops, "is_reference", False)
 
             for li, blk in enumerate(core.blocks):
                 def body(x, _blk=blk):
                     h = uma_ops.rms_norm(x, _blk.norm1, cfg.lmax)
+                    if use_rot:
+                        # fused rotation kernels (include/distmlip_hip.h):
+                        # rotate-gather per edge, split SO(2) GEMMs,
+                        # rotate-scatter per node over the dst CSR
+                        hf = h.float().contiguous()
+                        gatef = torch.sigmoid(
+                            _blk.edge_mlp(x_edge)[:, :cfg.lmax + 1])
+                        outs = []
+                        for (n0, n1, e0, e1) in ranges:
+                            def rchunk(hf_, gate_, _n0=n0, _n1=n1,
+                                       _e0=e0, _e1=e1, _b=_blk):
+                                Dc = D[_e0:_e1].float()
+                                xs = uma_ops._RotGather.apply(
+                                    hf_, Dc, pd.src[_e0:_e1])
+                                xd = uma_ops._RotGather.apply(
+                                    hf_, Dc, pd.dst[_e0:_e1])
+                                mt = uma_ops.so2_conv_split(
+                                    _b.msg, xs, xd, gate_[_e0:_e1], C)
+                                rp_loc = (pd.row_ptr[_n0:_n1 + 1]
+                                          - pd.row_ptr[_n0]).contiguous()
+                                dst_rel = (pd.dst[_e0:_e1].int()
+                                           - _n0).contiguous()
+                                return uma_ops._RotScatter.apply(
+                                    mt.float(), Dc, rp_loc, dst_rel)
+                            if len(ranges) > 1 and torch.is_grad_enabled():
+                                outs.append(torch.utils.checkpoint.checkpoint(
+                                    rchunk, hf, gatef, use_reentrant=False))
+                            else:
+                                outs.append(rchunk(hf, gatef))
+                        msg = torch.cat(outs, dim=0) if len(outs) > 1 \
+                            else outs[0]
+                        x2 = x + msg / cfg.avg_degree
+                        return x2 + uma_ops.node_ffn(_blk, cfg, x2)
                     if len(ranges) == 1:
                         hf = _flat(h).contiguous()
                         x_src = ops.gather(hf, pd.src,
@@ -215,7 +251,7 @@ class UmaSpmdEngine:
                         ).view(-1, S, C) / cfg.avg_degree
                     else:
                         acc = torch.zeros_like(x)
-                        for (e0, e1) in ranges:
+                        for (_n0, _n1, e0, e1) in ranges:
                             # nested checkpoint: without it the recompute
                             # backward holds EVERY chunk's [E_c, S, 2C]
                             # rotation graph at once (~20 GB x n_chunks)

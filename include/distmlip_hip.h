@@ -190,6 +190,25 @@ int dm_mace_symc_bwd_f32(const float* go, const float* x,
                          float* dx, int64_t N, int32_t C, int32_t T,
                          int32_t S_out, uint64_t stream);
 
+/* eSCN/UMA per-edge Wigner rotation family (round 2): replaces the
+ * per-edge D-matrix bmms of the reference's UMA path
+ * (escn_md.py:283-291 rotation setup; applications around every SO(2)
+ * conv).  D is [E, 81] row-major (S = 9, lmax 2); trans selects D^T
+ * (the inverse rotation).  rot_gather: out[e] = D_e . h[idx[e]]
+ * (idx NULL = identity rows).  rot_scatter: out[n] = (base[n]) +
+ * sum_{edges of n via row_ptr/perm} D_e . mt[e].  rot_dD: the D
+ * gradient, one [9,9] block per edge. */
+int dm_rot_gather_f32(const float* h, const int32_t* idx, const float* D,
+                      int32_t trans, float* out, int64_t E, int32_t C,
+                      uint64_t stream);
+int dm_rot_scatter_f32(const float* mt, const float* D, int32_t trans,
+                       const int32_t* perm, const int32_t* row_ptr,
+                       const float* base, float* out, int64_t N,
+                       int32_t C, uint64_t stream);
+int dm_rot_dD_f32(const float* go, const float* h, const int32_t* idx,
+                  int32_t trans, float* dD, int64_t E, int32_t C,
+                  uint64_t stream);
+
 const char* dm_hip_last_error(void);
 
 #ifdef __cplusplus
