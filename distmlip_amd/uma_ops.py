@@ -251,10 +251,12 @@ def so2_conv_split(conv: SO2Conv, xs: torch.Tensor, xd: torch.Tensor,
     lmax = conv.lmax
     m0, pm = m_indices(lmax)
     sp = _so2_split(conv)
-    out = xs.new_zeros(E, (lmax + 1) ** 2, C)
     ws, wd = sp["w0"]
     a0 = xs[:, m0, :].reshape(E, -1) @ ws.t() \
         + xd[:, m0, :].reshape(E, -1) @ wd.t()
+    # under bf16 autocast the GEMMs come back bf16: allocate the output
+    # in THAT dtype (indexed writes require matching dtypes)
+    out = xs.new_zeros(E, (lmax + 1) ** 2, C, dtype=a0.dtype)
     out[:, m0, :] = a0.view(E, lmax + 1, C) * gate[:, 0].view(E, 1, 1)
     for m in range(1, lmax + 1):
         plus, minus = pm[m - 1]
