@@ -70,8 +70,13 @@ def s2_act(x: torch.Tensor, cfg: UMAConfig,
     to_g, from_g = _grids(cfg, x)
 
     def one(t):
-        f = torch.nn.functional.silu(torch.einsum("gs,nsc->ngc", to_g, t))
-        return torch.einsum("sg,ngc->nsc", from_g, f)
+        # tensordot keeps these as ONE tall GEMM ([N*C, S] x [S, G]);
+        # the einsum forms lower to batch-N (S x G) bmm tiles (the
+        # rocBLAS ~1%-of-peak shape)
+        f = torch.tensordot(t, to_g, dims=([1], [1]))      # [N, C, G]
+        f = torch.nn.functional.silu(f)
+        out = torch.tensordot(f, from_g, dims=([2], [1]))  # [N, C, S]
+        return out.permute(0, 2, 1).contiguous()
 
     if x.shape[0] <= chunk:
         return one(x)
