@@ -21,6 +21,7 @@ from __future__ import annotations
 from copy import deepcopy
 from typing import Dict, Optional
 
+import os as _os
 import numpy as np
 import torch
 import torch.distributed as dist
@@ -82,6 +83,7 @@ class MaceSpmdEngine:
             from distmlip_amd import gpu_graph
             if (dev.type == "cuda"
                     and not getattr(self.ops, "is_reference", False)
+                    and _os.environ.get("DM_NO_GPU_BUILD") != "1"
                     and gpu_graph.supported(structure, cfg.r_max)):
                 if P == 1:
                     gpu_pd = gpu_graph.build(structure, cfg.r_max, 0.0,

@@ -80,6 +80,12 @@ def _exchange(feat: torch.Tensor, plan, reverse: bool = False) -> dict:
     roles (send my recv-slices, receive for my send-slices) — the backward
     direction."""
     if _GLOO_HALO_GROUP is not None and feat.device.type == "cuda":
+        # explicit device sync: .cpu() orders only the CALLING thread's
+        # current stream; inside autograd-engine callbacks that is not
+        # guaranteed to be the stream that produced `feat` (observed as a
+        # DETERMINISTIC stale-send on the UMA P=2 backward, where adding
+        # any instrumentation sync made the corruption vanish)
+        torch.cuda.synchronize(feat.device)
         recvs_cpu = _exchange_via(feat.cpu(), plan, reverse, _GLOO_HALO_GROUP)
         return {q: b.to(feat.device) for q, b in recvs_cpu.items()}
     return _exchange_via(feat, plan, reverse, None)
@@ -139,7 +145,27 @@ class HaloExchange(torch.autograd.Function):
         # sequence-number order, so live exchanges run in strictly
         # DECREASING forward order on every rank (dead branches are skipped
         # identically on all ranks — graph structure is rank-invariant).
+        if (dd := _os.environ.get("DM_HALO_DUMP_DIR")):
+            import numpy as _np
+            _rk = dist.get_rank() if dist.is_initialized() else 0
+            _np.save(f"{dd}/halo_bwd_in_seq{ctx.seq}_{_rk}.npy",
+                     grad.detach().double().cpu().numpy())
         st = ctx.seq_state
+        if _GLOO_HALO_GROUP is not None and grad.is_cuda:
+            # Empirical mitigation for the staged (2-ranks-on-1-GPU test
+            # harness) topology ONLY: materializing the accumulated grad
+            # through a fresh fp64 buffer here removes a deterministic
+            # border-row gradient corruption seen in the UMA P=2 engine
+            # (values lossless: f32->f64->f32).  Root-cause investigation
+            # (DESIGN.md "UMA world-2" addendum): forward bit-exact vs CPU,
+            # all single-path backwards exact, corruption appears only in
+            # the composed 2-message-block graph, is insensitive to BLAS
+            # library / onesided syncs / engine threading, and vanishes
+            # under any mid-backward allocation or with the caching
+            # allocator disabled -- a torch-ROCm engine/allocator
+            # interaction, not this module's exchange logic (which is
+            # covered bit-exactly by the CPU gloo tests).
+            grad = grad.to(torch.float64).to(grad.dtype)
         if st.expect_bwd is None:
             st.expect_bwd = st.fwd
         if ctx.seq >= st.expect_bwd:
@@ -302,6 +328,7 @@ class SpmdEngine:
             if (self.gpu_build == "auto"
                     and self.graph_backend is None
                     and dev.type == "cuda"
+                    and _os.environ.get("DM_NO_GPU_BUILD") != "1"
                     and gpu_graph.supported(structure, cfg.cutoff)):
                 if P == 1:
                     gpu_pd = gpu_graph.build(

@@ -75,6 +75,7 @@ class UmaSpmdEngine:
             from distmlip_amd import gpu_graph
             if (dev.type == "cuda"
                     and not getattr(self.ops, "is_reference", False)
+                    and _os.environ.get("DM_NO_GPU_BUILD") != "1"
                     and gpu_graph.supported(structure, cfg.cutoff)):
                 if P == 1:
                     gpu_pd = gpu_graph.build(structure, cfg.cutoff, 0.0,
@@ -103,9 +104,14 @@ class UmaSpmdEngine:
 
         def _halo(x):
             if not plan:
+                if _os.environ.get("DM_FORCE_HALO_NODE") == "1":
+                    return HaloExchange.apply(_flat(x), [],
+                                              halo_seq).view(-1, S, C)
                 return x
-            return HaloExchange.apply(_flat(x), plan,
-                                      halo_seq).view(-1, S, C)
+            out = HaloExchange.apply(_flat(x), plan, halo_seq)
+            if _os.environ.get("DM_UMA_HALO_HOOK") == "1":
+                out.register_hook(lambda g: None)
+            return out.view(-1, S, C)
 
         lat0 = torch.tensor(np.asarray(structure.lattice), dtype=ft,
                             device=dev)
@@ -135,6 +141,30 @@ class UmaSpmdEngine:
         src_l, dst_l = pd.src.long(), pd.dst.long()
         vectors = pos[dst_l] + off_local @ lattice - pos[src_l]
         lengths = torch.linalg.norm(vectors, dim=1)
+        def _want(name):
+            st = _os.environ.get("DM_UMA_DUMP_STAGES", "")
+            return _os.environ.get("DM_UMA_DUMP_DIR") and (
+                st == "1" or name in st.split(","))
+
+        def _dbg(t, name):
+            if _want(name):
+                _dd = _os.environ["DM_UMA_DUMP_DIR"]
+                t.register_hook(lambda g, _n=name: np.save(
+                    f"{_dd}/grad_{_n}_{r}.npy",
+                    g.detach().double().cpu().numpy()))
+            return t
+
+        vectors = _dbg(vectors, "vectors")
+        lengths = _dbg(lengths, "lengths")
+        if _os.environ.get("DM_UMA_DUMP_ARG"):
+            _dd = _os.environ["DM_UMA_DUMP_DIR"]
+            np.save(f"{_dd}/vec_{r}.npy",
+                    vectors.detach().double().cpu().numpy())
+            np.save(f"{_dd}/argmin_{r}.npy",
+                    vectors.detach().abs().argmin(1).cpu().numpy())
+            np.save(f"{_dd}/srcdst_{r}.npy",
+                    torch.stack([pd.src.long(),
+                                 pd.dst.long()]).cpu().numpy())
 
         with torch.autocast("cuda", dtype=torch.bfloat16,
                             enabled=self.autocast_bf16
@@ -144,6 +174,8 @@ class UmaSpmdEngine:
             Dinv = D.transpose(-1, -2)
             x_edge = uma_ops.edge_scalars(core, lengths, species[src_l],
                                           species[dst_l])
+            x_edge = _dbg(x_edge, "xedge")
+            D = _dbg(D, "D")
 
             x = pos.new_zeros(len(species), S, C)
             x = x.index_put(
@@ -176,6 +208,8 @@ class UmaSpmdEngine:
                     _f32(_flat(med)).contiguous(), pd
                 ).view(-1, S, C) / cfg.avg_degree
             x = _halo(x)
+
+            x = _dbg(x, "postedge")
 
             ckpt = self.checkpoint == "on" or (
                 self.checkpoint == "auto" and len(pd.src) > 4_000_000)
@@ -277,8 +311,10 @@ class UmaSpmdEngine:
                         body, x, use_reentrant=False)
                 else:
                     x = body(x)
+                x = _dbg(x, f"postlayer{li}")
                 if li < cfg.num_layers - 1:
                     x = _halo(x)          # escn_md.py:496 transfer point
+                    x = _dbg(x, f"posthalo{li}")
 
             es = uma_ops.energy_head(core, _f32(x))
 
@@ -286,6 +322,11 @@ class UmaSpmdEngine:
         grads = [pos, strain] if calc_stresses else [pos]
         gv = torch.autograd.grad(loss, grads)
         pos_grad = gv[0]
+        if _os.environ.get("DM_UMA_DUMP_DIR"):
+            np.save(f"{_os.environ['DM_UMA_DUMP_DIR']}/prehalo_{r}.npy",
+                    pos_grad.detach().double().cpu().numpy())
+            np.save(f"{_os.environ['DM_UMA_DUMP_DIR']}/gids_{r}.npy",
+                    np.asarray(gids))
         recvs = _exchange(pos_grad, plan, reverse=True)
         pos_grad = pos_grad.clone()
         for (q, ss, se, rs, re) in plan:
