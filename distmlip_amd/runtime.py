@@ -151,21 +151,6 @@ class HaloExchange(torch.autograd.Function):
             _np.save(f"{dd}/halo_bwd_in_seq{ctx.seq}_{_rk}.npy",
                      grad.detach().double().cpu().numpy())
         st = ctx.seq_state
-        if _GLOO_HALO_GROUP is not None and grad.is_cuda:
-            # Empirical mitigation for the staged (2-ranks-on-1-GPU test
-            # harness) topology ONLY: materializing the accumulated grad
-            # through a fresh fp64 buffer here removes a deterministic
-            # border-row gradient corruption seen in the UMA P=2 engine
-            # (values lossless: f32->f64->f32).  Root-cause investigation
-            # (DESIGN.md "UMA world-2" addendum): forward bit-exact vs CPU,
-            # all single-path backwards exact, corruption appears only in
-            # the composed 2-message-block graph, is insensitive to BLAS
-            # library / onesided syncs / engine threading, and vanishes
-            # under any mid-backward allocation or with the caching
-            # allocator disabled -- a torch-ROCm engine/allocator
-            # interaction, not this module's exchange logic (which is
-            # covered bit-exactly by the CPU gloo tests).
-            grad = grad.to(torch.float64).to(grad.dtype)
         if st.expect_bwd is None:
             st.expect_bwd = st.fwd
         if ctx.seq >= st.expect_bwd:
