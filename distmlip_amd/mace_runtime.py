@@ -177,13 +177,82 @@ class MaceSpmdEngine:
                 # holds the layer's OUTPUT
                 xd = {l: t for l, t in zip(_ls, xl)}
                 x_up = mace_ops.irreps_linear(_inter.linear_up, xd)
-                tp_w = _inter.radial(edge_feats).view(
-                    -1, len(_inter.paths), C)
-                gathered = {
-                    l: ops.gather(_flat(x_up[l]).contiguous(), pd.src,
-                                  csr=src_csr).view(-1, C, 2 * l + 1)
-                    for l in x_up}
-                if mace_ops.conv_tp_hip_available(_inter, Y, C):
+                use_hip = mace_ops.conv_tp_hip_available(_inter, Y, C)
+                E_tot = len(pd.src)
+                chunk_e = int(_os.environ.get("DM_MACE_CHUNK", 6_000_000))
+                if E_tot > chunk_e:
+                    # node-range-chunked message pass (the uma_runtime
+                    # pattern): per-edge transients ([E,P,C] radial
+                    # weights + 4 [E,d3,C] TP outputs) reach ~40 GB per
+                    # 6M edges, so big graphs stream dst-sorted
+                    # node-aligned ranges through per-chunk checkpoints
+                    rp_h = pd.row_ptr.long().cpu().numpy()
+                    ranges, n0 = [], 0
+                    for n in range(1, len(rp_h)):
+                        if rp_h[n] - rp_h[n0] >= chunk_e \
+                                or n == len(rp_h) - 1:
+                            ranges.append((n0, n, int(rp_h[n0]),
+                                           int(rp_h[n])))
+                            n0 = n
+                    flats = {l: _flat(x_up[l]).contiguous() for l in x_up}
+                    P = len(_inter.paths)
+
+                    def mchunk(ef_c, Y_c, *fl, _n0=0, _n1=0, _e0=0, _e1=0):
+                        tp_w = _inter.radial(ef_c).view(-1, P, C)
+                        src_c = pd.src[_e0:_e1].long()
+                        g = {l: f[src_c].view(-1, C, 2 * l + 1)
+                             for l, f in zip(sorted(x_up), fl)}
+                        rp_rel = (pd.row_ptr[_n0:_n1 + 1]
+                                  - pd.row_ptr[_n0]).contiguous()
+                        dst_rel = (pd.dst[_e0:_e1]
+                                   - pd.dst.new_tensor(_n0)).contiguous()
+                        outs = []
+                        if use_hip:
+                            mts = mace_ops.conv_tp_hip(
+                                _inter, g[0][:, :, 0], g.get(1), Y_c,
+                                tp_w)
+                            for l3 in range(4):
+                                outs.append(ops.scatter_rows(
+                                    _flat(mts[l3]).contiguous(), dst_rel,
+                                    rp_rel, _n1 - _n0))
+                        else:
+                            msgs = mace_ops.conv_tp_messages(
+                                _inter, g, Y_c, tp_w)
+                            for l3 in sorted(msgs):
+                                outs.append(ops.scatter_rows(
+                                    _flat(msgs[l3]).contiguous(), dst_rel,
+                                    rp_rel, _n1 - _n0))
+                        return tuple(outs)
+
+                    parts = []
+                    for (a, b, e0, e1) in ranges:
+                        kw = dict(_n0=a, _n1=b, _e0=e0, _e1=e1)
+                        args = (edge_feats[e0:e1], Y[e0:e1],
+                                *[flats[l] for l in sorted(x_up)])
+                        if torch.is_grad_enabled():
+                            parts.append(torch.utils.checkpoint.checkpoint(
+                                lambda *t, _kw=kw: mchunk(*t, **_kw),
+                                *args, use_reentrant=False))
+                        else:
+                            parts.append(mchunk(*args, **kw))
+                    m = {}
+                    if use_hip:
+                        for l3 in range(4):
+                            cat = torch.cat([p_[l3] for p_ in parts], 0)
+                            m[l3] = cat.view(-1, 2 * l3 + 1, C).permute(
+                                0, 2, 1).contiguous()
+                    else:
+                        l3s = sorted({pp[2] for pp in _inter.paths})
+                        for i, l3 in enumerate(l3s):
+                            cat = torch.cat([p_[i] for p_ in parts], 0)
+                            m[l3] = cat.view(-1, C, 2 * l3 + 1)
+                elif use_hip:
+                    tp_w = _inter.radial(edge_feats).view(
+                        -1, len(_inter.paths), C)
+                    gathered = {
+                        l: ops.gather(_flat(x_up[l]).contiguous(), pd.src,
+                                      csr=src_csr).view(-1, C, 2 * l + 1)
+                        for l in x_up}
                     # fused per-edge TP kernel (one wave per edge,
                     # include/distmlip_hip.h); outputs are [E, d3, C]
                     # l-major — transpose back after the scatter ([N,*]
@@ -199,6 +268,12 @@ class MaceSpmdEngine:
                         ).view(-1, d3, C)
                         m[l3] = sc.permute(0, 2, 1).contiguous()
                 else:
+                    tp_w = _inter.radial(edge_feats).view(
+                        -1, len(_inter.paths), C)
+                    gathered = {
+                        l: ops.gather(_flat(x_up[l]).contiguous(), pd.src,
+                                      csr=src_csr).view(-1, C, 2 * l + 1)
+                        for l in x_up}
                     msgs = mace_ops.conv_tp_messages(_inter, gathered, Y,
                                                      tp_w)
                     m = {l3: ops.scatter_edges(

@@ -554,6 +554,12 @@ class HipOps:
     def scatter_edges(self, msg, pd, base=None):
         return _SegSum.apply(msg, pd.dst, pd.row_ptr, pd.n_atoms, base)
 
+    def scatter_rows(self, msg, dst_rel, row_ptr_rel, n_rows):
+        """Segment-sum of dst-sorted msg rows over an arbitrary local CSR
+        (node-range-chunked message passes; dst_rel/row_ptr_rel are
+        relative to the range start)."""
+        return _SegSum.apply(msg, dst_rel, row_ptr_rel, n_rows, None)
+
     # -- raw (non-differentiable) primitives for the hand-sequenced conv
     #    backward (distmlip_amd.conv); see ops_base docstring ------------
 

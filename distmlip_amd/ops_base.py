@@ -79,6 +79,9 @@ class OpsBackend(Protocol):
 
     def scatter_edges(self, msg, pd, base=None) -> torch.Tensor: ...
 
+    def scatter_rows(self, msg, dst_rel, row_ptr_rel,
+                     n_rows) -> torch.Tensor: ...
+
     def r_gather_add3(self, zs, zd, ze, pd): ...
 
     def r_combine_fwd(self, cg, w, base) -> torch.Tensor: ...

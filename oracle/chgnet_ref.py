@@ -222,6 +222,9 @@ class CpuRefOps:
     def scatter_edges(self, msg, pd, base=None):
         return self._scatter(msg, pd.dst, pd.n_atoms, base)
 
+    def scatter_rows(self, msg, dst_rel, row_ptr_rel, n_rows):
+        return self._scatter(msg, dst_rel.long(), n_rows, None)
+
     def scatter_lines(self, msg, pd, base=None):
         return self._scatter(msg, pd.l_dst, pd.n_bonds, base)
 
