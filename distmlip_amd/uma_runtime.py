@@ -193,9 +193,11 @@ class UmaSpmdEngine:
                 return t.float() if ac else t
 
             E_all = len(pd.src)
-            if E_all > 2_000_000:
+            deg_chunk = int(_os.environ.get("DM_UMA_DEG_CHUNK",
+                                            2_000_000))
+            if E_all > deg_chunk:
                 accd = torch.zeros_like(x)
-                step_e = 2_000_000
+                step_e = deg_chunk
                 for e0 in range(0, E_all, step_e):
                     e1 = min(e0 + step_e, E_all)
                     md = uma_ops.edge_degree_embed(core, x_edge[e0:e1],

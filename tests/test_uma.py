@@ -210,6 +210,24 @@ def test_uma_spmd_gloo(tmp_path):
     assert dF < 5e-9, f"UMA SPMD force error {dF}"
 
 
+def test_uma_edge_degree_chunked_matches(monkeypatch):
+    """The chunked edge-degree scatter (the >2M-edge memory path) must
+    match the single-pass scatter exactly."""
+    from distmlip_amd.uma_runtime import UmaSpmdEngine
+
+    s = diamond_si((6, 2, 2), jitter=0.1, seed=4)
+    s.species = np.asarray(s.species) % 3
+    core = _small_core(seed=9)
+    ref = UmaSpmdEngine(core, world=1, threads=2, device="cpu",
+                        ops=CpuRefOps()).step(s)
+    monkeypatch.setenv("DM_UMA_DEG_CHUNK", "700")
+    got = UmaSpmdEngine(core, world=1, threads=2, device="cpu",
+                        ops=CpuRefOps()).step(s)
+    assert abs(ref["energy"].item() - got["energy"].item()) < 1e-10
+    dF = (ref["forces_owned"] - got["forces_owned"]).abs().max().item()
+    assert dF < 1e-10, dF
+
+
 def test_uma_engine_chunked_matches(monkeypatch):
     """The node-range-chunked message pass (the >1.5M-edge memory path)
     must match the single-chunk path exactly."""
