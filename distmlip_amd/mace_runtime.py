@@ -163,8 +163,14 @@ class MaceSpmdEngine:
         x = {0: core.node_embedding[species].unsqueeze(-1)}
         es_sum = None
 
+        chunk_e = int(_os.environ.get("DM_MACE_CHUNK", 6_000_000))
+        # when the chunked message pass is active its per-chunk
+        # checkpoints already bound the [E,*,C] transients; an OUTER
+        # body checkpoint on top would recompute every chunk forward a
+        # second time (3x total) for node-level savings only
         ckpt = self.checkpoint == "on" or (
-            self.checkpoint == "auto" and len(pd.src) > 4_000_000)
+            self.checkpoint == "auto"
+            and 4_000_000 < len(pd.src) <= chunk_e)
 
         for i, (inter, prod) in enumerate(zip(core.interactions,
                                               core.products)):
@@ -179,7 +185,6 @@ class MaceSpmdEngine:
                 x_up = mace_ops.irreps_linear(_inter.linear_up, xd)
                 use_hip = mace_ops.conv_tp_hip_available(_inter, Y, C)
                 E_tot = len(pd.src)
-                chunk_e = int(_os.environ.get("DM_MACE_CHUNK", 6_000_000))
                 if E_tot > chunk_e:
                     # node-range-chunked message pass (the uma_runtime
                     # pattern): per-edge transients ([E,P,C] radial
