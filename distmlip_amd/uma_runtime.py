@@ -213,15 +213,22 @@ class UmaSpmdEngine:
 
             x = _dbg(x, "postedge")
 
+            chunk_edges = int(_os.environ.get("DM_UMA_CHUNK",
+                                               1_500_000))
+            # when the chunked message pass is active its per-chunk
+            # checkpoints already bound the [E,S,2C] rotation
+            # transients; an outer per-layer checkpoint on top would
+            # recompute every chunk forward a second time (mirrors the
+            # MACE engine; measured there: 3.03 -> 2.30 s/step at 512k)
             ckpt = self.checkpoint == "on" or (
-                self.checkpoint == "auto" and len(pd.src) > 4_000_000)
+                self.checkpoint == "auto"
+                and 4_000_000 < len(pd.src) <= chunk_edges)
             src_csr = (pd.src_perm, pd.src_row_ptr) \
                 if hasattr(pd, "src_perm") else None
 
             # per-edge rotated tensors ([E, S, 2C]) reach ~100 GB at 11M
             # edges — the message pass runs over contiguous node-range
             # chunks (dst-sorted edges), bounding transients
-            chunk_edges = int(_os.environ.get("DM_UMA_CHUNK", 1_500_000))
             E_tot = len(pd.src)
             N_loc = x.shape[0]
             if E_tot > chunk_edges:

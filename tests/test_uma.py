@@ -228,6 +228,31 @@ def test_uma_edge_degree_chunked_matches(monkeypatch):
     assert dF < 1e-10, dF
 
 
+def test_uma_spmd_chunked_gloo(tmp_path, monkeypatch):
+    """Chunked message pass composed with SPMD halos (world=2): the
+    per-rank node ranges cover owned+ghost rows, so chunk boundaries
+    interact with the halo regions — pin it against the fp64 oracle."""
+    monkeypatch.setenv("DM_UMA_CHUNK", "700")
+    monkeypatch.setenv("DM_UMA_DEG_CHUNK", "900")
+    world = 2
+    init_file = str(tmp_path / "pg_init")
+    mp.spawn(_worker, args=(world, init_file, str(tmp_path)),
+             nprocs=world, join=True)
+    s = diamond_si((12, 2, 2), jitter=0.12, seed=3)
+    s.species = np.asarray(s.species) % 3
+    core = _small_core(seed=9)
+    src, dst, off = _graph(s)
+    ref = uma_oracle_forward(core, s, src, dst, off)
+    F = np.zeros((s.num_atoms, 3))
+    for r in range(world):
+        E_r = np.load(f"{tmp_path}/E_{r}.npy")[0]
+        assert abs(E_r - ref["energy"].item()) < 1e-9
+        gids = np.load(f"{tmp_path}/gids_{r}.npy")
+        F[gids] = np.load(f"{tmp_path}/F_{r}.npy")
+    dF = np.abs(F - ref["forces"].numpy()).max()
+    assert dF < 5e-9, f"chunked SPMD force error {dF}"
+
+
 def test_uma_engine_chunked_matches(monkeypatch):
     """The node-range-chunked message pass (the >1.5M-edge memory path)
     must match the single-chunk path exactly."""
