@@ -41,6 +41,26 @@ def test_mace_engine_gpu_vs_oracle():
 
 
 @requires_gpu
+def test_mace_engine_gpu_chunked_matches(monkeypatch):
+    """The node-range-chunked fused message pass (the >6M-edge memory
+    path, forced small here) must match the unchunked fused path."""
+    from distmlip_amd.mace_model import MACEConfig, MACECore
+    from distmlip_amd.mace_runtime import MaceSpmdEngine
+    from distmlip_amd.structures import diamond_si
+
+    s = diamond_si((6, 2, 2), jitter=0.1, seed=3)
+    s.species = np.asarray(s.species) % 3
+    cfg = MACEConfig(n_elements=3, channels=64)
+    core = MACECore.seeded(cfg, seed=1).float()
+    ref = MaceSpmdEngine(core, world=1, threads=4).step(s)
+    monkeypatch.setenv("DM_MACE_CHUNK", "3000")
+    got = MaceSpmdEngine(core, world=1, threads=4).step(s)
+    assert abs(ref["energy"].item() - got["energy"].item()) < 1e-5
+    dF = (ref["forces_owned"] - got["forces_owned"]).abs().max().item()
+    assert dF < 1e-4, dF
+
+
+@requires_gpu
 def test_mace_engine_gpu_checkpointed_matches():
     """checkpoint='on' (the big-workload path) must match checkpoint-off
     bit-for-bit on the same inputs (same kernels, same order)."""

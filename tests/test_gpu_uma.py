@@ -39,6 +39,29 @@ def test_uma_engine_gpu_vs_oracle():
 
 
 @requires_gpu
+def test_uma_engine_gpu_chunked_matches(monkeypatch):
+    """The node-range-chunked rotation-kernel message pass (forced small
+    chunks) must match the single-chunk path."""
+    from distmlip_amd.structures import diamond_si
+    from distmlip_amd.uma_model import UMAConfig, UMACore
+    from distmlip_amd.uma_runtime import UmaSpmdEngine
+
+    s = diamond_si((6, 2, 2), jitter=0.2, seed=3)
+    s.species = np.asarray(s.species) % 3
+    cfg = UMAConfig(n_elements=3, sphere_channels=64, num_layers=2)
+    core = UMACore.seeded(cfg, seed=1).float()
+    ref = UmaSpmdEngine(core, world=1, threads=4).step(s)
+    monkeypatch.setenv("DM_UMA_CHUNK", "3000")
+    monkeypatch.setenv("DM_UMA_DEG_CHUNK", "5000")
+    got = UmaSpmdEngine(core, world=1, threads=4).step(s)
+    assert abs(ref["energy"].item() - got["energy"].item()) < 1e-4 * max(
+        1.0, abs(ref["energy"].item()))
+    scale = ref["forces_owned"].abs().max().item()
+    dF = (ref["forces_owned"] - got["forces_owned"]).abs().max().item()
+    assert dF < 1e-3 * max(1.0, scale), (dF, scale)
+
+
+@requires_gpu
 def test_uma_engine_bf16_autocast_close():
     """The bf16-autocast bench mode stays within bf16-resolution of the
     fp32 engine on the same inputs (no silent divergence)."""
