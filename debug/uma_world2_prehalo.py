@@ -40,15 +40,19 @@ def worker(rank, world, init_file, out_dir, dev_mode, env):
             dist.destroy_process_group()
 
 if __name__ == "__main__":
-    import pathlib
+    # NOTE: the DM_ABL graph-ablation levers used mid-investigation were
+    # stripped from the engine once the bisection concluded (DESIGN.md
+    # §13 addendum 2); the remaining levers are the dump hooks below
+    # plus DM_FORCE_HALO_NODE / DM_HALO_DUMP_DIR in the runtimes and
+    # AMD_SERIALIZE_KERNEL=3 (which makes the run exact).
     out = os.environ.get("DM_PREHALO_OUT", "/tmp/umaph")
     os.makedirs(out, exist_ok=True)
-    B = "nonorm,noffn,nohalo1"
     TRIALS = [
-        ("arg", {"DM_ABL": B, "DM_UMA_DUMP_ARG": "1"}),
-        ("hv", {"DM_ABL": B, "DM_UMA_DUMP_STAGES": "vectors"}),
-        ("hD", {"DM_ABL": B, "DM_UMA_DUMP_STAGES": "D"}),
-        ("hx", {"DM_ABL": B, "DM_UMA_DUMP_STAGES": "xedge"}),
+        ("arg", {"DM_UMA_DUMP_ARG": "1"}),
+        ("hv", {"DM_UMA_DUMP_STAGES": "vectors"}),
+        ("hD", {"DM_UMA_DUMP_STAGES": "D"}),
+        ("hx", {"DM_UMA_DUMP_STAGES": "xedge"}),
+        ("serial", {"AMD_SERIALIZE_KERNEL": "3"}),
     ]
     for abl, env in TRIALS:
         cn, gn = f"cpu_{abl}", f"gpu_{abl}"
